@@ -1,0 +1,322 @@
+"""Communication backends: RCCL-native core + torch.distributed fallback.
+
+Plays the role Horovod's C++ core plays for the reference
+(reference distributed_optimizer.py:21-26 imports allreduce_async_,
+synchronize, broadcast_async_, broadcast, init, rank, size, local_size,
+local_rank from horovod.torch.mpi_ops) — same module-level API surface,
+MI355X-native underneath:
+
+- ``RcclCoreBackend``: our C++/HIP extension (``mgx_comm``) driving RCCL
+  directly — one communicator per job, ncclAllReduce on a dedicated
+  high-priority non-blocking HIP stream, hipEvent-based handles so
+  ``synchronize`` is a device-side stream-wait, not a host stall.
+  Bootstrap: the ncclUniqueId is exchanged over a torch.distributed gloo
+  group (env-var rendezvous, torchrun-compatible; no MPI anywhere).
+- ``TorchDistBackend``: torch.distributed collectives — gloo for the
+  CPU-only test tier (BASELINE config 1), and the ``nccl`` backend (which
+  IS RCCL on ROCm) as a GPU fallback / A-B reference for the native core.
+
+Process bootstrap is env-var rendezvous (RANK / WORLD_SIZE / MASTER_ADDR /
+MASTER_PORT), replacing the reference's ``mpirun + hvd.init()``
+(reference dist_mpi.sh:12, dist_trainer.py:133).
+"""
+from __future__ import annotations
+
+import os
+import datetime
+
+import torch
+import torch.distributed as dist
+
+from .. import settings
+from ..settings import logger
+
+__all__ = [
+    'init', 'shutdown', 'initialized', 'rank', 'size', 'local_rank',
+    'local_size', 'allreduce_async_', 'synchronize', 'broadcast',
+    'broadcast_async_', 'barrier', 'backend_name',
+]
+
+_backend = None
+
+
+class Handle:
+    """Async-op handle. ``wait()`` orders the CALLER's stream (or host on
+    CPU) after the collective; mirrors Horovod's handle+synchronize."""
+
+    def wait(self):
+        raise NotImplementedError
+
+
+class _TorchWorkHandle(Handle):
+    def __init__(self, work, tensor, post_scale=None):
+        self._work = work
+        self._tensor = tensor
+        self._post_scale = post_scale
+        self._done = False
+
+    def wait(self):
+        if self._done:
+            return self._tensor
+        self._work.wait()   # on NCCL: current stream waits on comm stream
+        if self._post_scale is not None:
+            self._tensor.mul_(self._post_scale)
+        self._done = True
+        return self._tensor
+
+
+class _NoopHandle(Handle):
+    def __init__(self, tensor):
+        self._tensor = tensor
+
+    def wait(self):
+        return self._tensor
+
+
+class TorchDistBackend:
+    """torch.distributed-based backend (gloo on CPU, nccl/RCCL on GPU)."""
+
+    name = 'torch'
+
+    def __init__(self, backend=None, device=None):
+        if not dist.is_initialized():
+            world_size = int(os.environ.get('WORLD_SIZE', '1'))
+            if world_size > 1 or 'MASTER_ADDR' in os.environ:
+                if backend is None:
+                    backend = 'nccl' if torch.cuda.is_available() else 'gloo'
+                os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+                os.environ.setdefault('MASTER_PORT', '29500')
+                dist.init_process_group(
+                    backend=backend,
+                    timeout=datetime.timedelta(seconds=300))
+        self._initialized_pg = dist.is_initialized()
+        self._backend = dist.get_backend() if self._initialized_pg else None
+        self._supports_avg = (self._backend == 'nccl')
+
+    def rank(self):
+        return dist.get_rank() if self._initialized_pg else 0
+
+    def size(self):
+        return dist.get_world_size() if self._initialized_pg else 1
+
+    def local_rank(self):
+        return int(os.environ.get('LOCAL_RANK', self.rank()))
+
+    def local_size(self):
+        return int(os.environ.get('LOCAL_WORLD_SIZE', self.size()))
+
+    def allreduce_async(self, tensor, average=True, name=None):
+        if self.size() == 1:
+            return _NoopHandle(tensor)
+        if average and self._supports_avg:
+            work = dist.all_reduce(tensor, op=dist.ReduceOp.AVG, async_op=True)
+            return _TorchWorkHandle(work, tensor)
+        work = dist.all_reduce(tensor, op=dist.ReduceOp.SUM, async_op=True)
+        scale = (1.0 / self.size()) if average else None
+        return _TorchWorkHandle(work, tensor, post_scale=scale)
+
+    def broadcast_async(self, tensor, root):
+        if self.size() == 1:
+            return _NoopHandle(tensor)
+        work = dist.broadcast(tensor, src=root, async_op=True)
+        return _TorchWorkHandle(work, tensor)
+
+    def broadcast(self, tensor, root):
+        if self.size() > 1:
+            dist.broadcast(tensor, src=root)
+        return tensor
+
+    def barrier(self):
+        if self.size() > 1:
+            dist.barrier()
+
+    def shutdown(self):
+        if self._initialized_pg and dist.is_initialized():
+            dist.destroy_process_group()
+
+
+class _RcclCoreHandle(Handle):
+    def __init__(self, core, hid, tensor):
+        self._core = core
+        self._hid = hid
+        self._tensor = tensor
+        self._done = False
+
+    def wait(self):
+        if not self._done:
+            # device-side: current torch stream waits on the comm stream's
+            # hipEvent recorded after ncclAllReduce (no host stall)
+            self._core.wait_handle(self._hid,
+                                   torch.cuda.current_stream().cuda_stream)
+            self._done = True
+        return self._tensor
+
+
+class RcclCoreBackend:
+    """Native RCCL comm core (C++ extension mgx_comm).
+
+    Dedicated non-blocking high-priority HIP stream for collectives,
+    hipEvent handles; the ncclUniqueId is exchanged through a gloo group.
+    """
+
+    name = 'rccl'
+
+    def __init__(self):
+        from ..kernels import load_comm_core
+        self._core = load_comm_core()   # raises if extension missing on GPU
+        world_size = int(os.environ.get('WORLD_SIZE', '1'))
+        rk = int(os.environ.get('RANK', '0'))
+        if not dist.is_initialized() and world_size > 1:
+            os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+            os.environ.setdefault('MASTER_PORT', '29500')
+            dist.init_process_group(backend='gloo',
+                                    timeout=datetime.timedelta(seconds=300))
+        self._rank = rk
+        self._size = world_size
+        dev = torch.device('cuda', int(os.environ.get('LOCAL_RANK', rk % max(torch.cuda.device_count(), 1))))
+        torch.cuda.set_device(dev)
+        if world_size > 1:
+            if rk == 0:
+                uid = self._core.unique_id()
+                t = torch.ByteTensor(list(uid))
+            else:
+                t = torch.zeros(self._core.unique_id_size(), dtype=torch.uint8)
+            dist.broadcast(t, src=0)
+            uid_bytes = bytes(t.tolist())
+        else:
+            uid_bytes = self._core.unique_id()
+        self._core.init(self._rank, self._size, uid_bytes)
+        logger.info('[rank %d] RCCL comm core initialized (world=%d)',
+                    self._rank, self._size)
+
+    def rank(self):
+        return self._rank
+
+    def size(self):
+        return self._size
+
+    def local_rank(self):
+        return int(os.environ.get('LOCAL_RANK', self._rank))
+
+    def local_size(self):
+        return int(os.environ.get('LOCAL_WORLD_SIZE', self._size))
+
+    def allreduce_async(self, tensor, average=True, name=None):
+        if self._size == 1:
+            return _NoopHandle(tensor)
+        hid = self._core.allreduce_async(
+            tensor, average, torch.cuda.current_stream().cuda_stream)
+        return _RcclCoreHandle(self._core, hid, tensor)
+
+    def broadcast_async(self, tensor, root):
+        if self._size == 1:
+            return _NoopHandle(tensor)
+        hid = self._core.broadcast_async(
+            tensor, root, torch.cuda.current_stream().cuda_stream)
+        return _RcclCoreHandle(self._core, hid, tensor)
+
+    def broadcast(self, tensor, root):
+        h = self.broadcast_async(tensor, root)
+        h.wait()
+        return tensor
+
+    def barrier(self):
+        if dist.is_initialized():
+            dist.barrier()
+        torch.cuda.synchronize()
+
+    def shutdown(self):
+        self._core.destroy()
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def init(backend='auto'):
+    """Initialize the communication backend.
+
+    backend: 'auto' | 'rccl' (native core) | 'nccl' | 'gloo' | 'torch'.
+    'auto' picks the native RCCL core on GPU (falling back to
+    torch.distributed nccl if the extension is unavailable) and gloo on CPU.
+    """
+    global _backend
+    if _backend is not None:
+        return _backend
+    want = os.environ.get('MGX_COMM_BACKEND', backend)
+    if want == 'auto':
+        if torch.cuda.is_available():
+            try:
+                _backend = RcclCoreBackend()
+            except Exception as e:  # extension missing -> torch-dist nccl
+                logger.warning('RCCL core unavailable (%s); falling back to '
+                               'torch.distributed nccl', e)
+                _backend = TorchDistBackend(backend='nccl')
+        else:
+            _backend = TorchDistBackend(backend='gloo')
+    elif want == 'rccl':
+        _backend = RcclCoreBackend()
+    elif want in ('nccl', 'gloo'):
+        _backend = TorchDistBackend(backend=want)
+    else:
+        _backend = TorchDistBackend()
+    return _backend
+
+
+def initialized():
+    return _backend is not None
+
+
+def _get():
+    global _backend
+    if _backend is None:
+        init()
+    return _backend
+
+
+def shutdown():
+    global _backend
+    if _backend is not None:
+        _backend.shutdown()
+        _backend = None
+
+
+def backend_name():
+    return _get().name
+
+
+def rank():
+    return _get().rank()
+
+
+def size():
+    return _get().size()
+
+
+def local_rank():
+    return _get().local_rank()
+
+
+def local_size():
+    return _get().local_size()
+
+
+def allreduce_async_(tensor, average=True, name=None):
+    """Async in-place all-reduce; returns a Handle (Horovod-API parity:
+    reference distributed_optimizer.py:339)."""
+    return _get().allreduce_async(tensor, average=average, name=name)
+
+
+def synchronize(handle):
+    """Wait an async handle; returns the reduced tensor."""
+    return handle.wait()
+
+
+def broadcast(tensor, root_rank=0, name=None):
+    return _get().broadcast(tensor, root_rank)
+
+
+def broadcast_async_(tensor, root_rank=0, name=None):
+    return _get().broadcast_async(tensor, root_rank)
+
+
+def barrier():
+    _get().barrier()
