@@ -1,0 +1,659 @@
+"""DLTrainer — training runtime (reference dl_trainer.py:138-1007).
+
+Same surface: ``DLTrainer(rank, nworkers, dist=..., batch_size=...,
+is_weak_scaling=..., dataset=..., dnn=..., lr=...)`` with ``train()``,
+``test()``, ``update_model()``, ``zero_grad()``, ``update_nworker()``,
+``save_checkpoint`` / ``load_model_from_file``, the per-dataset SGD
+config with bn/bias no-decay split (reference :216-248), the LR-schedule
+family (reference :578-709), and the headline log line
+``Time per iteration including communication: %f, Speed: %f images/s``
+(reference :981-984).
+
+MI355X-first differences:
+- bf16 autocast (torch.amp on ROCm) replaces apex AMP O2 (reference
+  :278-281); fp32 master weights and fp32 gradients.
+- torch.nn.CTCLoss replaces warp-ctc (reference :214-215).
+- synthetic datasets are first-class (no dataset download on the GPU
+  box; BASELINE runs are synthetic/random-init); on GPU the synthetic
+  pipeline keeps batches device-resident (zero H2D in steady state).
+- no torch.nn.DataParallel path: scaling is one process per GPU over
+  RCCL (the reference's ngpus>1 DataParallel is vestigial).
+"""
+from __future__ import annotations
+
+import os
+import time
+
+import torch
+import torch.nn as nn
+import torch.optim as optim
+from torch.utils.data import DataLoader
+from torch.utils.data.distributed import DistributedSampler
+
+from . import models
+from . import settings
+from . import data as mgx_data
+from .settings import logger
+
+if torch.cuda.is_available():
+    torch.backends.cudnn.benchmark = True   # MIOpen autotune on ROCm
+
+_support_dnns = ['resnet20', 'resnet32', 'resnet44', 'resnet56',
+                 'resnet110', 'resnet_mod20', 'resnet_mod32',
+                 'resnet_mod44', 'resnet_mod56', 'resnet_mod110',
+                 'preresnet20', 'preresnet32', 'preresnet44', 'preresnet56',
+                 'preresnet110', 'resnet18', 'resnet34', 'resnet50',
+                 'resnet101', 'resnet152', 'resnext29_8_64',
+                 'resnext29_16_64', 'densenet100_12', 'densenet121',
+                 'densenet161', 'densenet201', 'vgg16', 'vgg19', 'vgg16i',
+                 'alexnet', 'googlenet', 'inceptionv3', 'inceptionv4',
+                 'caffe_cifar', 'lenet', 'mnistnet', 'fcn5net', 'lr',
+                 'lstm', 'lstman4']
+_support_datasets = ['imagenet', 'cifar10', 'mnist', 'ptb', 'an4']
+
+NUM_CLASSES = {'imagenet': 1000, 'cifar10': 10, 'mnist': 10}
+
+
+def create_net(num_classes, dnn='resnet20', **kwargs):
+    """Model factory (reference dl_trainer.py:87-135) — all entries
+    native (no torchvision)."""
+    ext = None
+    if dnn == 'lstman4':
+        net, ext = models.LSTMAN4(datapath=kwargs.get('datapath'))
+    elif dnn == 'lstm':
+        net = models.lstm(vocab_size=kwargs.get('vocab_size', 10000),
+                          batch_size=kwargs.get('batch_size', 20))
+    elif dnn == 'mnistnet':
+        net = models.MnistNet()
+    elif dnn == 'fcn5net':
+        net = models.FCN5Net()
+    elif dnn == 'lenet':
+        net = models.LeNet()
+    elif dnn == 'lr':
+        net = models.LinearRegression()
+    elif dnn == 'vgg16':
+        net = models.VGG('VGG16', num_classes=num_classes)
+    elif dnn == 'vgg19':
+        net = models.VGG('VGG19', num_classes=num_classes)
+    elif dnn in models.__dict__:
+        net = models.__dict__[dnn](num_classes=num_classes)
+    else:
+        errstr = 'Unsupported neural network %s' % dnn
+        logger.error(errstr)
+        raise ValueError(errstr)
+    return net, ext
+
+
+class DLTrainer:
+    def __init__(self, rank, size, master=None, dist=True, ngpus=1,
+                 batch_size=32, is_weak_scaling=True, data_dir='./data',
+                 dataset='cifar10', dnn='resnet20', lr=0.1, nworkers=1,
+                 prefix=None, sparsity=0.95, pretrain=None, num_steps=35,
+                 tb_writer=None, dtype=None, synthetic=None):
+        self.rank = rank
+        self.pretrain = pretrain
+        self.dataset = dataset
+        self.prefix = prefix
+        self.num_steps = num_steps
+        self.ngpus = ngpus
+        self.writer = tb_writer
+        self.nworkers = nworkers
+        if is_weak_scaling:
+            self.batch_size = batch_size        # per-worker batch fixed
+        else:
+            self.batch_size = max(batch_size // max(nworkers, 1), 1)
+        self.is_cuda = ngpus > 0 and torch.cuda.is_available()
+        self.data_dir = data_dir
+        # synthetic data unless a real dataset directory exists
+        if synthetic is None:
+            synthetic = not (data_dir and os.path.isdir(data_dir)
+                             and data_dir not in ('', './data'))
+        self.synthetic = synthetic
+        self.dnn = dnn
+        self.lr = lr
+        self.base_lr = lr
+        self.m = 0.9
+        # bf16 autocast on GPU by default (BASELINE dtype contract)
+        if dtype is None:
+            dtype = os.environ.get('MGX_DTYPE',
+                                   'bf16' if self.is_cuda else 'fp32')
+        self.compute_dtype = {'fp32': torch.float32,
+                              'bf16': torch.bfloat16,
+                              'fp16': torch.float16}[dtype]
+        self.autocast_enabled = (self.is_cuda
+                                 and self.compute_dtype != torch.float32
+                                 and dnn not in ('lstm', 'lstman4'))
+
+        if self.is_cuda:
+            torch.cuda.set_device(rank % max(torch.cuda.device_count(), 1))
+        self.device = (torch.device('cuda')
+                       if self.is_cuda else torch.device('cpu'))
+
+        self.ext = None
+        self.data_prepare()
+        kwargs = {'datapath': data_dir, 'vocab_size': self.vocab_size,
+                  'batch_size': self.batch_size}
+        self.net, self.ext = create_net(self.num_classes, dnn, **kwargs)
+        self.net = self.net.to(self.device)
+        if settings.DEBUG and rank == 0:
+            logger.info('%s: %d parameters', dnn,
+                        sum(p.numel() for p in self.net.parameters()))
+
+        # loss
+        if dataset == 'an4':
+            self.criterion = nn.CTCLoss(blank=0, reduction='sum',
+                                        zero_infinity=True)
+        else:
+            self.criterion = nn.CrossEntropyLoss().to(self.device)
+
+        # per-dataset SGD hyperparameters (reference :216-248)
+        weight_decay = 1e-4
+        self.m = 0.9
+        nesterov = False
+        if dataset == 'an4':
+            self.lstman4_lr_epoch_tag = 0
+        elif dataset == 'ptb':
+            self.m = 0
+            weight_decay = 0
+        elif dataset == 'imagenet':
+            self.m = 0.875
+            weight_decay = 2 * 3.0517578125e-05
+        decay, no_decay = [], []
+        for name, param in self.net.named_parameters():
+            if not param.requires_grad:
+                continue
+            if len(param.shape) == 1 or 'bn' in name or 'bias' in name:
+                no_decay.append(param)
+            else:
+                decay.append(param)
+        parameters = [{'params': no_decay, 'weight_decay': 0.},
+                      {'params': decay, 'weight_decay': weight_decay}]
+        self.optimizer = optim.SGD(parameters, lr=self.lr, momentum=self.m,
+                                   weight_decay=weight_decay,
+                                   nesterov=nesterov)
+
+        self.train_epoch = 0
+        self.train_iter = 0
+        self.avg_loss_per_epoch = 0.0
+        self.accuracy = 0.0
+        self.loss = 0.0
+        self.times = []
+        self.display = 40
+        self.io_time = 0.0
+        self.forward_time = 0.0
+        self.backward_time = 0.0
+
+        if self.pretrain is not None and os.path.isfile(self.pretrain):
+            self.load_model_from_file(self.pretrain)
+
+    # ------------------------------------------------------------------
+    # data
+    # ------------------------------------------------------------------
+    def data_prepare(self):
+        self.vocab_size = 10000
+        self.num_classes = NUM_CLASSES.get(self.dataset, 10)
+        self._gpu_pool = None
+        ds = None
+        test_ds = None
+        if self.dataset == 'imagenet':
+            if self.synthetic:
+                ds = mgx_data.synthetic_imagenet(
+                    length=1280 * max(self.nworkers, 1) * 8)
+            else:
+                from .data.hdf5 import DatasetHDF5
+                ds = DatasetHDF5(os.path.join(self.data_dir,
+                                              'imagenet-shuffled.hdf5'),
+                                 'train')
+                test_ds = DatasetHDF5(
+                    os.path.join(self.data_dir, 'imagenet-shuffled.hdf5'),
+                    'val')
+        elif self.dataset == 'cifar10':
+            ds = mgx_data.synthetic_cifar10()
+        elif self.dataset == 'mnist':
+            ds = mgx_data.synthetic_mnist()
+        elif self.dataset == 'ptb':
+            if self.synthetic:
+                ds = mgx_data.SyntheticPTBDataset(
+                    vocab_size=self.vocab_size, num_steps=self.num_steps)
+            else:
+                train, valid, _, vocab = mgx_data.ptb_raw_data(self.data_dir)
+                self.vocab_size = vocab
+                ds = mgx_data.PTBDataset(train, self.batch_size,
+                                         self.num_steps)
+                test_ds = mgx_data.PTBDataset(valid, self.batch_size,
+                                              self.num_steps)
+        elif self.dataset == 'an4':
+            if self.synthetic:
+                ds = mgx_data.SyntheticAudioDataset()
+            else:
+                from .models.deepspeech import LABELS
+                ds = mgx_data.SpectrogramDataset(
+                    {}, os.path.join(self.data_dir,
+                                     'an4_train_manifest.csv'), LABELS)
+                test_ds = mgx_data.SpectrogramDataset(
+                    {}, os.path.join(self.data_dir, 'an4_val_manifest.csv'),
+                    LABELS)
+        else:
+            raise ValueError('unknown dataset %s' % self.dataset)
+
+        self.trainset = ds
+        self.testset = test_ds if test_ds is not None else ds
+        sampler = None
+        if self.nworkers > 1 and self.dataset != 'an4':
+            sampler = DistributedSampler(ds, num_replicas=self.nworkers,
+                                         rank=self.rank)
+        self.train_sampler = sampler
+        collate = mgx_data.an4_collate if self.dataset == 'an4' else None
+        if self.dataset == 'an4' and self.nworkers > 1:
+            batch_sampler = mgx_data.DistributedBucketingSampler(
+                ds, batch_size=self.batch_size,
+                num_replicas=self.nworkers, rank=self.rank)
+            self.trainloader = DataLoader(ds, batch_sampler=batch_sampler,
+                                          collate_fn=collate, num_workers=2)
+            self.batch_sampler = batch_sampler
+        else:
+            self.batch_sampler = None
+            self.trainloader = DataLoader(
+                ds, batch_size=self.batch_size,
+                shuffle=(sampler is None), sampler=sampler,
+                collate_fn=collate, num_workers=2, drop_last=True)
+        self.testloader = DataLoader(self.testset,
+                                     batch_size=self.batch_size,
+                                     shuffle=False, collate_fn=collate,
+                                     num_workers=1)
+        self.num_batches_per_epoch = max(
+            len(self.trainset) // (self.batch_size * max(self.nworkers, 1)),
+            1)
+        self.data_iterator = None
+
+    def data_iter(self):
+        """Restartable iterator (reference dl_trainer.py:568-576)."""
+        if self.data_iterator is None:
+            self.data_iterator = iter(self.trainloader)
+        try:
+            return next(self.data_iterator)
+        except StopIteration:
+            if self.train_sampler is not None:
+                self.train_sampler.set_epoch(self.train_epoch + 1)
+            if self.batch_sampler is not None:
+                self.batch_sampler.set_epoch(self.train_epoch + 1)
+            self.data_iterator = iter(self.trainloader)
+            return next(self.data_iterator)
+
+    def fetch_data(self):
+        """One (inputs, labels) batch on the right device.
+
+        On GPU synthetic image workloads, batches come from a resident
+        device pool (no H2D, no host dataloader in the timed region)."""
+        if self._gpu_pool is None and self.is_cuda and self.synthetic \
+                and self.dataset in ('imagenet', 'cifar10', 'mnist'):
+            shape = {'imagenet': (3, 224, 224), 'cifar10': (3, 32, 32),
+                     'mnist': (1, 28, 28)}[self.dataset]
+            self._gpu_pool = mgx_data.GPUBatchPool.images(
+                self.batch_size, shape, self.num_classes, self.device,
+                n_batches=8, seed=self.rank)
+        if self._gpu_pool is not None:
+            return self._gpu_pool.next()
+        batch = self.data_iter()
+        if self.dataset == 'an4':
+            inputs, targets, input_sizes, target_sizes = batch
+            inputs = inputs.to(self.device, non_blocking=True)
+            return (inputs, input_sizes), (targets, target_sizes)
+        inputs, labels = batch
+        inputs = inputs.to(self.device, non_blocking=True)
+        labels = labels.to(self.device, non_blocking=True)
+        return inputs, labels
+
+    # ------------------------------------------------------------------
+    # checkpoints (the reference constructs the filename but never calls
+    # torch.save — implemented here)
+    # ------------------------------------------------------------------
+    def get_checkpoint_path(self, weights_dir='./weights'):
+        d = os.path.join(
+            weights_dir, self.prefix or 'default',
+            '%s-n%d-bs%d-lr%.4f' % (self.dnn, self.nworkers,
+                                    self.batch_size, self.base_lr))
+        os.makedirs(d, exist_ok=True)
+        return os.path.join(d, '%s-rank%d-epoch%d.pth'
+                            % (self.dnn, self.rank, self.train_epoch))
+
+    def save_checkpoint(self, state=None, filename=None):
+        if filename is None:
+            filename = self.get_checkpoint_path()
+        if state is None:
+            state = {'iter': self.train_iter, 'epoch': self.train_epoch,
+                     'state': self.net.state_dict(),
+                     'optim': self.optimizer.state_dict()}
+        torch.save(state, filename)
+        logger.info('checkpoint saved: %s', filename)
+        return filename
+
+    def load_model_from_file(self, filename):
+        checkpoint = torch.load(filename, map_location=self.device,
+                                weights_only=False)
+        self.net.load_state_dict(checkpoint['state'])
+        self.train_epoch = checkpoint['epoch']
+        self.train_iter = checkpoint['iter']
+        if 'optim' in checkpoint:
+            try:
+                self.optimizer.load_state_dict(checkpoint['optim'])
+            except Exception as e:
+                logger.warning('optimizer state not restored: %s', e)
+        logger.info('resumed from %s (epoch %d iter %d)', filename,
+                    self.train_epoch, self.train_iter)
+
+    # ------------------------------------------------------------------
+    # elastic re-rank (reference dl_trainer.py:545-566)
+    # ------------------------------------------------------------------
+    def update_nworker(self, nworkers, new_rank=-1):
+        if new_rank >= 0:
+            self.rank = new_rank
+        self.nworkers = nworkers
+        self.data_prepare()   # rebuild samplers/loaders for the new world
+
+    # ------------------------------------------------------------------
+    # train / test
+    # ------------------------------------------------------------------
+    def zero_grad(self):
+        self.optimizer.zero_grad()
+
+    def update_model(self):
+        self.optimizer.step()
+
+    def update_optimizer(self, optimizer):
+        self.optimizer = optimizer
+
+    def _forward_loss(self, inputs, labels, hidden=None):
+        if self.dnn == 'lstm':
+            outputs, hidden = self.net(inputs, hidden)
+            loss = self.criterion(
+                outputs.contiguous().view(-1, self.net.ntokens),
+                labels.contiguous().view(-1))
+            return loss, outputs, hidden
+        if self.dnn == 'lstman4':
+            x, input_sizes = inputs
+            targets, target_sizes = labels
+            outputs, output_sizes = self.net(x, input_sizes)
+            loss = self.ctc_loss(outputs, targets, output_sizes,
+                                 target_sizes)
+            loss = loss / x.size(0)
+            return loss, outputs, None
+        outputs = self.net(inputs)
+        loss = self.criterion(outputs, labels)
+        return loss, outputs, None
+
+    def ctc_loss(self, logits, targets, output_sizes, target_sizes):
+        """CTC over (T, N, C) raw logits via torch.nn.CTCLoss (replaces
+        warp-ctc, reference dl_trainer.py:214-215)."""
+        log_probs = torch.nn.functional.log_softmax(logits.float(), dim=-1)
+        return self.criterion(log_probs, targets,
+                              output_sizes.to(torch.long),
+                              target_sizes.to(torch.long))
+
+    def train(self, num_of_iters=1, data=None, hidden=None):
+        self.loss = 0.0
+        s = time.time()
+        for _ in range(num_of_iters):
+            self.adjust_learning_rate(self.train_epoch, self.optimizer)
+            if self.train_iter % self.num_batches_per_epoch == 0 \
+                    and self.train_iter > 0:
+                self.train_epoch += 1
+                if self.rank == 0:
+                    logger.info('epoch %d done (lr %.6f)', self.train_epoch,
+                                self.lr)
+
+            ss = time.time()
+            if data is None:
+                inputs, labels = self.fetch_data()
+            else:
+                inputs, labels = data
+            if self.dnn == 'lstm' and hidden is not None:
+                hidden = models.repackage_hidden(hidden)
+            self.io_time += time.time() - ss
+
+            sf = time.time()
+            with torch.autocast(device_type='cuda',
+                                dtype=self.compute_dtype,
+                                enabled=self.autocast_enabled):
+                loss, outputs, hidden = self._forward_loss(inputs, labels,
+                                                           hidden)
+            self.forward_time += time.time() - sf
+
+            sb = time.time()
+            loss.backward()
+            self.backward_time += time.time() - sb
+
+            loss_value = loss.item()
+            self.loss += loss_value
+            self.avg_loss_per_epoch += loss_value
+            self.train_iter += 1
+        self.times.append(time.time() - s)
+
+        if self.train_iter % self.display == 0 and self.rank == 0:
+            n = min(len(self.times), self.display)
+            avg = sum(self.times[-n:]) / n
+            throughput = self.batch_size * self.nworkers / avg
+            logger.info(
+                'Time per iteration including communication: %f, Speed: '
+                '%f images/s', avg, throughput)
+            logger.info('  phase times (last %d iters): io %.4f fwd %.4f '
+                        'bwd %.4f', n, self.io_time / self.display,
+                        self.forward_time / self.display,
+                        self.backward_time / self.display)
+            self.io_time = self.forward_time = self.backward_time = 0.0
+        return self.loss / num_of_iters, hidden
+
+    def test(self, epoch=0):
+        self.net.eval()
+        correct1 = correct5 = total = 0
+        total_loss = 0.0
+        total_wer = 0.0
+        nbatches = 0
+        decoder = (models.GreedyDecoder() if self.dataset == 'an4'
+                   else None)
+        with torch.no_grad():
+            for batch in self.testloader:
+                if self.dataset == 'an4':
+                    inputs, targets, input_sizes, target_sizes = batch
+                    inputs = inputs.to(self.device)
+                    outputs, output_sizes = self.net(inputs, input_sizes)
+                    decoded = decoder.decode(
+                        outputs.permute(1, 0, 2).cpu(), output_sizes)
+                    # reconstruct target strings
+                    off = 0
+                    for i, ts in enumerate(target_sizes):
+                        t = targets[off:off + int(ts)]
+                        off += int(ts)
+                        tstr = ''.join(decoder.int2char[int(c)] for c in t)
+                        total_wer += decoder.wer(decoded[i], tstr)
+                        nbatches += 1
+                    continue
+                if self.dnn == 'lstm':
+                    x, y = batch
+                    x, y = x.to(self.device), y.to(self.device)
+                    hidden = self.net.init_hidden()
+                    if x.size(0) != self.net.batch_size:
+                        continue
+                    out, _ = self.net(x, hidden)
+                    loss = self.criterion(
+                        out.contiguous().view(-1, self.net.ntokens),
+                        y.contiguous().view(-1))
+                    total_loss += loss.item()
+                    nbatches += 1
+                    continue
+                x, y = batch
+                x, y = x.to(self.device), y.to(self.device)
+                out = self.net(x)
+                _, pred = out.topk(5, dim=1)
+                correct = pred.eq(y.view(-1, 1))
+                correct1 += correct[:, 0].sum().item()
+                correct5 += correct.sum().item()
+                total += y.size(0)
+                nbatches += 1
+        self.net.train()
+        if self.dataset == 'an4':
+            wer = total_wer / max(nbatches, 1)
+            logger.info('epoch %d test WER: %.4f', epoch, wer)
+            return wer
+        if self.dnn == 'lstm':
+            import math
+            ppl = math.exp(total_loss / max(nbatches, 1))
+            logger.info('epoch %d test perplexity: %.4f', epoch, ppl)
+            return ppl
+        acc1 = correct1 / max(total, 1)
+        acc5 = correct5 / max(total, 1)
+        logger.info('epoch %d test top-1 %.4f top-5 %.4f', epoch, acc1,
+                    acc5)
+        self.accuracy = acc1
+        return acc1
+
+    # ------------------------------------------------------------------
+    # LR schedules (reference dl_trainer.py:578-709)
+    # ------------------------------------------------------------------
+    def _adjust_learning_rate_lstman4(self, progress, optimizer):
+        if self.lstman4_lr_epoch_tag != progress:
+            self.lstman4_lr_epoch_tag = progress
+            self.lr = self.lr / 1.01
+            for g in optimizer.param_groups:
+                g['lr'] = self.lr
+        return self.lr
+
+    def _adjust_learning_rate_lstmptb(self, progress, optimizer):
+        first, second, third = 63, 60, 80
+        if progress < first:
+            lr = self.base_lr
+        elif progress < second:
+            lr = self.base_lr * 0.1
+        elif progress < third:
+            lr = self.base_lr * 0.01
+        else:
+            lr = self.base_lr * 0.001
+        self.lr = lr
+        for g in optimizer.param_groups:
+            g['lr'] = self.lr
+        return self.lr
+
+    def _adjust_learning_rate_general(self, progress, optimizer):
+        warmup = 5
+        if settings.WARMUP and progress < warmup:
+            warmup_total_iters = self.num_batches_per_epoch * warmup
+            min_lr = self.base_lr / warmup_total_iters
+            lr_interval = (self.base_lr - min_lr) / warmup_total_iters
+            self.lr = min_lr + lr_interval * self.train_iter
+            for g in optimizer.param_groups:
+                g['lr'] = self.lr
+            return self.lr
+        first, second, third = 81, 122, 155
+        if self.dataset == 'imagenet':
+            first, second, third = 30, 60, 80
+        elif self.dataset == 'ptb':
+            first, second, third = 24, 60, 80
+        if progress < first:
+            lr = self.base_lr
+        elif progress < second:
+            lr = self.base_lr * 0.1
+        elif progress < third:
+            lr = self.base_lr * 0.01
+        else:
+            lr = self.base_lr * 0.001
+        self.lr = lr
+        for g in optimizer.param_groups:
+            g['lr'] = self.lr
+        return self.lr
+
+    def _adjust_learning_rate_vgg16(self, progress, optimizer):
+        if progress > 0 and progress % 25 == 0:
+            self.lr = self.base_lr / (2 ** (progress // 25))
+        for g in optimizer.param_groups:
+            g['lr'] = self.lr
+        return self.lr
+
+    def _adjust_learning_rate_customized(self, progress, optimizer):
+        warmup = 10
+        npe = self.num_batches_per_epoch
+        if settings.WARMUP and progress < warmup:
+            total_iters = warmup * npe
+            min_lr = self.base_lr / total_iters
+            lr_interval = (self.base_lr - min_lr) / total_iters
+            self.lr = min_lr + lr_interval * self.train_iter
+        elif progress < 15:
+            self.lr = self.base_lr
+        elif progress < 25:
+            self.lr = self.base_lr * 0.1
+        elif progress < 35:
+            self.lr = self.base_lr * 0.01
+        else:
+            self.lr = self.base_lr * 0.001
+        for g in optimizer.param_groups:
+            g['lr'] = self.lr
+        return self.lr
+
+    def _adjust_learning_rate_cosine(self, progress, optimizer):
+        import math
+        warmup = 14
+        max_epochs = 40
+        npe = self.num_batches_per_epoch
+        if settings.WARMUP and progress < warmup:
+            total_iters = warmup * npe
+            min_lr = self.base_lr / total_iters
+            lr_interval = (self.base_lr - min_lr) / total_iters
+            self.lr = min_lr + lr_interval * self.train_iter
+        elif progress < max_epochs:
+            e = progress - warmup
+            es = max_epochs - warmup
+            self.lr = 0.5 * (1 + math.cos(math.pi * e / es)) * self.base_lr
+        for g in optimizer.param_groups:
+            g['lr'] = self.lr
+        return self.lr
+
+    def adjust_learning_rate(self, progress, optimizer):
+        if self.dnn == 'lstman4':
+            return self._adjust_learning_rate_lstman4(
+                self.train_iter // self.num_batches_per_epoch, optimizer)
+        if self.dnn == 'lstm':
+            return self._adjust_learning_rate_lstmptb(progress, optimizer)
+        if self.dnn in ('vgg16', 'vgg19'):
+            return self._adjust_learning_rate_vgg16(progress, optimizer)
+        return self._adjust_learning_rate_general(progress, optimizer)
+
+    def finish(self):
+        if self.writer is not None:
+            self.writer.close()
+
+
+def train_with_single(dnn, dataset, data_dir, nworkers, lr, batch_size,
+                      nsteps_update, max_epochs, num_steps=1,
+                      save_epochs=0):
+    """Single-GPU training entry (reference dl_trainer.py:956-1007) —
+    the 1-GPU point of the scaling curve."""
+    from .profiling import benchmark
+    torch.cuda.set_device(0) if torch.cuda.is_available() else None
+    trainer = DLTrainer(0, 1, dist=False, batch_size=batch_size,
+                        is_weak_scaling=True, ngpus=1, data_dir=data_dir,
+                        dataset=dataset, dnn=dnn, lr=lr, nworkers=1,
+                        prefix='singlegpu', num_steps=num_steps)
+    seq_layernames, layerwise_times, layerwise_sizes = benchmark(trainer)
+    logger.info('layerwise backward times (sum): %f s',
+                sum(layerwise_times))
+    iters_per_epoch = trainer.num_batches_per_epoch
+    times = []
+    display = 40
+    hidden = trainer.net.init_hidden() if dnn == 'lstm' else None
+    for epoch in range(max_epochs):
+        if trainer.train_iter >= iters_per_epoch * max_epochs:
+            break
+        for i in range(iters_per_epoch):
+            s = time.time()
+            trainer.zero_grad()
+            for _ in range(nsteps_update):
+                _, hidden = trainer.train(1, hidden=hidden)
+            trainer.update_model()
+            times.append(time.time() - s)
+            if i % display == 0 and i > 0:
+                avg = sum(times[-display:]) / min(len(times), display)
+                logger.info('Time per iteration including communication: '
+                            '%f, Speed: %f images/s', avg,
+                            batch_size * nsteps_update / avg)
+        if save_epochs and (epoch + 1) % save_epochs == 0:
+            trainer.save_checkpoint()
+    return trainer
