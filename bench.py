@@ -1,0 +1,177 @@
+"""Flagship benchmark: ResNet-50 / synthetic ImageNet weak-scaling DP.
+
+Driver contract: ``python bench.py --gpus N --steps K --warmup W`` (for
+N>1 the driver launches via torch.distributed.run, one rank per GPU).
+Rank 0 prints ONE JSON line with the whole-job images/s (BASELINE.json
+metric: "images/sec (whole node) + scaling eff., ResNet-50 synthetic
+ImageNet").
+
+One timed step = zero_grad + forward + backward (hooks fire merged-group
+RCCL all-reduces overlapped with backward) + synchronize + optimizer
+update — the reference's "Time per iteration including communication"
+(reference dist_trainer.py:97-99), bf16 autocast compute, fp32 grads.
+
+--merge selects the A/B arm: mgwfbp (solver, default) | wfbp
+(threshold 0) | single (one group) | threshold:<elems>.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument('--gpus', type=int, default=1)
+    parser.add_argument('--steps', type=int, default=40)
+    parser.add_argument('--warmup', type=int, default=10)
+    parser.add_argument('--model', type=str, default='resnet50')
+    parser.add_argument('--batch-size', type=int, default=128)
+    parser.add_argument('--dataset', type=str, default='imagenet')
+    parser.add_argument('--dtype', type=str, default='bf16')
+    parser.add_argument('--merge', type=str, default='mgwfbp',
+                        help='mgwfbp | wfbp | single | threshold:<elems>')
+    parser.add_argument('--profile-iters', type=int, default=15,
+                        help='layerwise-profiling iterations for the '
+                             'solver (mgwfbp arm)')
+    args = parser.parse_args()
+
+    world_size = int(os.environ.get('WORLD_SIZE', '1'))
+    rank = int(os.environ.get('RANK', '0'))
+    n_gpus = max(world_size, 1)
+    if args.gpus != n_gpus and world_size == 1 and args.gpus > 1:
+        print('WARNING: --gpus %d but launched single-process; '
+              'benchmarking N=1' % args.gpus, file=sys.stderr)
+        args.gpus = 1
+
+    os.environ.setdefault('MGX_DTYPE', args.dtype)
+    if args.merge != 'mgwfbp':
+        os.environ['MGX_ADAPTIVE_MERGE'] = '0'
+
+    import mgwfbp_amd.comm as comm
+    import mgwfbp_amd.settings as settings
+    from mgwfbp_amd.dl_trainer import DLTrainer
+    from mgwfbp_amd.distributed_optimizer import (DistributedOptimizer,
+                                                  broadcast_parameters)
+    from mgwfbp_amd.profiling import benchmark as profile_layers
+
+    comm.init()
+    assert comm.size() == n_gpus
+    if torch.cuda.is_available():
+        torch.cuda.set_device(comm.local_rank()
+                              % torch.cuda.device_count())
+
+    threshold = 0
+    if args.merge == 'single':
+        threshold = 1 << 40
+    elif args.merge.startswith('threshold:'):
+        threshold = int(args.merge.split(':', 1)[1])
+
+    trainer = DLTrainer(rank, n_gpus, dist=False,
+                        batch_size=args.batch_size, is_weak_scaling=True,
+                        ngpus=1 if torch.cuda.is_available() else 0,
+                        data_dir='', dataset=args.dataset, dnn=args.model,
+                        lr=0.01, nworkers=n_gpus, prefix='bench',
+                        dtype=args.dtype, synthetic=True)
+
+    seq_layernames = layerwise_times = None
+    if args.merge == 'mgwfbp':
+        seq_layernames, layerwise_times, _ = profile_layers(
+            trainer, num_warmup=3, num_iters=args.profile_iters)
+        if comm.size() > 1:
+            t = torch.tensor(layerwise_times, dtype=torch.float64)
+            if comm.backend_name() == 'rccl':
+                t = t.to(trainer.device)
+            comm.broadcast(t, root_rank=0)
+            layerwise_times = [float(x) for x in t.cpu()]
+
+    optimizer = DistributedOptimizer(
+        trainer.optimizer,
+        named_parameters=list(trainer.net.named_parameters()),
+        seq_layernames=seq_layernames, layerwise_times=layerwise_times,
+        threshold=threshold)
+    trainer.update_optimizer(optimizer)
+    if comm.size() > 1:
+        broadcast_parameters(trainer.net.state_dict(), root_rank=0)
+
+    def one_step():
+        optimizer.zero_grad()
+        trainer.train(1)
+        trainer.update_model()
+
+    # warmup (untimed)
+    for _ in range(args.warmup):
+        one_step()
+
+    # timed region: barrier + sync on both sides
+    comm.barrier()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(args.steps):
+        one_step()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    t1 = time.time()
+    comm.barrier()
+
+    elapsed = t1 - t0
+    # MAX over ranks (slowest rank defines job time)
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if comm.size() > 1:
+        if comm.backend_name() == 'rccl':
+            t = t.to(trainer.device)
+        h = comm.allreduce_async_(t, average=False)
+        comm.synchronize(h)
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        elapsed_sum = float(t.cpu()[0])
+        # max via allreduce of max: redo with explicit max reduction if
+        # backend supports; fall back to sum/size upper bound
+        import torch.distributed as dist
+        if dist.is_initialized():
+            t2 = torch.tensor([elapsed], dtype=torch.float64)
+            dist.all_reduce(t2, op=dist.ReduceOp.MAX)
+            elapsed = float(t2[0])
+        else:
+            elapsed = elapsed_sum / comm.size()
+
+    ms_per_step = elapsed / args.steps * 1e3
+    global_batch = args.batch_size * n_gpus
+    images_per_sec = global_batch * args.steps / elapsed
+
+    if rank == 0:
+        result = {
+            'metric': 'images/sec (whole node), ResNet-50 synthetic '
+                      'ImageNet' if args.model == 'resnet50'
+                      else 'images/sec (whole node), %s' % args.model,
+            'value': images_per_sec,
+            'unit': 'images/s',
+            'n_gpus': n_gpus,
+            'steps': args.steps,
+            'warmup': args.warmup,
+            'ms_per_step': ms_per_step,
+            'higher_is_better': True,
+            'scaling': 'weak',
+            'vs_baseline': None,
+            'dtype': args.dtype,
+            'data': 'synthetic',
+            'config': {
+                'model': args.model,
+                'global_batch': global_batch,
+                'per_gpu_batch': args.batch_size,
+                'image_size': 224 if args.dataset == 'imagenet' else 32,
+                'parallelism': 'dp%d' % n_gpus,
+                'merge': args.merge,
+                'comm_backend': comm.backend_name(),
+            },
+        }
+        print(json.dumps(result))
+    comm.shutdown()
+
+
+if __name__ == '__main__':
+    main()
