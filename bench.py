@@ -58,8 +58,15 @@ def main():
                                                   broadcast_parameters)
     from mgwfbp_amd.profiling import benchmark as profile_layers
 
+    t_start = time.time()
+
+    def phase(msg):
+        print('[bench +%6.1fs] %s' % (time.time() - t_start, msg),
+              file=sys.stderr, flush=True)
+
     comm.init()
     assert comm.size() == n_gpus
+    phase('comm initialized (%s)' % comm.backend_name())
     if torch.cuda.is_available():
         torch.cuda.set_device(comm.local_rank()
                               % torch.cuda.device_count())
@@ -77,6 +84,8 @@ def main():
                         lr=0.01, nworkers=n_gpus, prefix='bench',
                         dtype=args.dtype, synthetic=True)
 
+    phase('trainer built (%s/%s bs%d)' % (args.model, args.dataset,
+                                          args.batch_size))
     seq_layernames = layerwise_times = None
     if args.merge == 'mgwfbp':
         seq_layernames, layerwise_times, _ = profile_layers(
@@ -88,6 +97,7 @@ def main():
             comm.broadcast(t, root_rank=0)
             layerwise_times = [float(x) for x in t.cpu()]
 
+    phase('layerwise profile done')
     optimizer = DistributedOptimizer(
         trainer.optimizer,
         named_parameters=list(trainer.net.named_parameters()),
@@ -102,9 +112,11 @@ def main():
         trainer.train(1)
         trainer.update_model()
 
+    phase('optimizer ready')
     # warmup (untimed)
     for _ in range(args.warmup):
         one_step()
+    phase('warmup done (%d steps)' % args.warmup)
 
     # timed region: barrier + sync on both sides
     comm.barrier()

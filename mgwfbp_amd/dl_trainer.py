@@ -35,8 +35,13 @@ from . import settings
 from . import data as mgx_data
 from .settings import logger
 
+# MIOpen autotune: FAST find mode keeps fresh-box startup bounded (the
+# default exhaustive search costs minutes per new conv config); flip
+# MGX_MIOPEN_BENCHMARK=1 for the exhaustive-search arm.
+os.environ.setdefault('MIOPEN_FIND_MODE', 'FAST')
 if torch.cuda.is_available():
-    torch.backends.cudnn.benchmark = True   # MIOpen autotune on ROCm
+    torch.backends.cudnn.benchmark = \
+        os.environ.get('MGX_MIOPEN_BENCHMARK', '0') == '1'
 
 _support_dnns = ['resnet20', 'resnet32', 'resnet44', 'resnet56',
                  'resnet110', 'resnet_mod20', 'resnet_mod32',
@@ -382,6 +387,12 @@ class DLTrainer:
         loss = self.criterion(outputs, labels)
         return loss, outputs, None
 
+    def autocast(self):
+        """Autocast context used for every forward (bf16 on GPU)."""
+        return torch.autocast(device_type='cuda',
+                              dtype=self.compute_dtype,
+                              enabled=self.autocast_enabled)
+
     def ctc_loss(self, logits, targets, output_sizes, target_sizes):
         """CTC over (T, N, C) raw logits via torch.nn.CTCLoss (replaces
         warp-ctc, reference dl_trainer.py:214-215)."""
@@ -412,9 +423,7 @@ class DLTrainer:
             self.io_time += time.time() - ss
 
             sf = time.time()
-            with torch.autocast(device_type='cuda',
-                                dtype=self.compute_dtype,
-                                enabled=self.autocast_enabled):
+            with self.autocast():
                 loss, outputs, hidden = self._forward_loss(inputs, labels,
                                                            hidden)
             self.forward_time += time.time() - sf

@@ -125,18 +125,22 @@ def benchmark(trainer, num_warmup=5, num_iters=50):
             if torch.cuda.is_available():
                 torch.cuda.synchronize()
         trainer.net.zero_grad(set_to_none=False)
-        if trainer.dnn == 'lstm':
-            outputs, hidden = trainer.net(inputs, hidden)
-            loss = trainer.criterion(
-                outputs.contiguous().view(-1, trainer.net.ntokens),
-                labels.contiguous().view(-1))
-        elif trainer.dnn == 'lstman4':
-            outputs, output_sizes = trainer.net(inputs[0], inputs[1])
-            loss = trainer.ctc_loss(outputs, labels[0], output_sizes,
-                                    labels[1])
-        else:
-            outputs = trainer.net(inputs)
-            loss = trainer.criterion(outputs, labels)
+        # profile under the SAME autocast the training loop uses so the
+        # solver's layer times (and MIOpen's tuned configs) match the
+        # dtype that actually runs
+        with trainer.autocast():
+            if trainer.dnn == 'lstm':
+                outputs, hidden = trainer.net(inputs, hidden)
+                loss = trainer.criterion(
+                    outputs.contiguous().view(-1, trainer.net.ntokens),
+                    labels.contiguous().view(-1))
+            elif trainer.dnn == 'lstman4':
+                outputs, output_sizes = trainer.net(inputs[0], inputs[1])
+                loss = trainer.ctc_loss(outputs, labels[0], output_sizes,
+                                        labels[1])
+            else:
+                outputs = trainer.net(inputs)
+                loss = trainer.criterion(outputs, labels)
         if i >= num_warmup:
             p.start()
         loss.backward()

@@ -160,15 +160,22 @@ class TestOptimizerGPU:
             named_parameters=list(net_b.named_parameters()), threshold=0)
         assert opt_b._fused_sgd is not None, 'HIP fused SGD must be live'
         crit = nn.CrossEntropyLoss()
+        params_a = dict(net_a.named_parameters())
+        params_b = dict(net_b.named_parameters())
         for step in range(3):
             g = torch.Generator().manual_seed(step)
             x = torch.randn(8, 3, 32, 32, generator=g).cuda()
             y = torch.randint(0, 10, (8,), generator=g).cuda()
             opt_a.zero_grad()
             crit(net_a(x), y).backward()
-            opt_a.step()
+            # conv backward on ROCm is not bit-deterministic run-to-run;
+            # copy net_a's grads into net_b's bucket views so this test
+            # isolates the OPTIMIZER path, not MIOpen numerics
             opt_b.zero_grad()
-            crit(net_b(x), y).backward()
+            with torch.no_grad():
+                for name, pa in params_a.items():
+                    params_b[name].grad.copy_(pa.grad)
+            opt_a.step()
             opt_b.step()
         torch.cuda.synchronize()
         for pa, pb in zip(net_a.parameters(), net_b.parameters()):
