@@ -136,12 +136,14 @@ class GPUBatchPool:
 
     @classmethod
     def images(cls, batch_size, shape, num_classes, device, n_batches=8,
-               dtype=torch.float32, seed=0):
+               dtype=torch.float32, seed=0, channels_last=False):
         g = torch.Generator(device='cpu').manual_seed(seed)
         batches = []
         for _ in range(n_batches):
             x = torch.randn((batch_size,) + tuple(shape), generator=g)
             y = torch.randint(0, num_classes, (batch_size,), generator=g)
-            batches.append((x.to(device=device, dtype=dtype),
-                            y.to(device)))
+            x = x.to(device=device, dtype=dtype)
+            if channels_last and x.dim() == 4:
+                x = x.to(memory_format=torch.channels_last)
+            batches.append((x, y.to(device)))
         return cls(batches)

@@ -38,6 +38,7 @@ import torch
 from . import comm
 from . import settings
 from . import solver
+from . import utils
 from .settings import logger
 
 _GROUP_PAD = 64   # pad each layer's offset to 64 elements (16B*dtype align
@@ -197,16 +198,19 @@ class _DistributedOptimizer(torch.optim.Optimizer):
             self._merged_parameters[new_key] = flat
             self._merged_parameter_names[flat] = new_key
             self._merged_parameter_offsets[new_key] = offsets
-            # zero-copy: grads become views into the flat buffer
+            # zero-copy: grads become views into the flat buffer, in each
+            # parameter's NATIVE memory format (channels_last weights get
+            # NHWC accumulation — no transposes on the hot path)
             for k, o in zip(group, offsets):
                 p = self._named_parameters[k]
-                p.grad = flat[o:o + p.numel()].view_as(p)
+                p.grad = utils.grad_view_like(flat[o:o + p.numel()], p)
             if self._comm_dtype != torch.float32:
                 self._group_comm_buffers[new_key] = torch.zeros(
                     total, dtype=self._comm_dtype, device=first_p.device)
                 if self._use_hip:
                     from . import kernels as _k
-                    grads = [self._named_parameters[k].grad for k in group]
+                    grads = [utils.dense_flat_alias(
+                        self._named_parameters[k].grad) for k in group]
                     self._group_pack_tables[new_key] = _k.PackTable(
                         grads, offsets)
 
@@ -230,14 +234,15 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                                   # groups: fall back to the torch loop
                 if not p.requires_grad or p.grad is None:
                     continue
-                params.append(p.data)
-                grads.append(p.grad.data)
+                params.append(utils.dense_flat_alias(p.data))
+                grads.append(utils.dense_flat_alias(p.grad.data))
                 wds.append(g.get('weight_decay', 0.0))
                 if momentum != 0.0:
                     state = self.state[p]
                     if 'momentum_buffer' not in state:
                         state['momentum_buffer'] = torch.zeros_like(p.data)
-                    momenta.append(state['momentum_buffer'])
+                    momenta.append(
+                        utils.dense_flat_alias(state['momentum_buffer']))
         if not params:
             return None
         from . import kernels as _k
@@ -400,7 +405,10 @@ def broadcast_parameters(params, root_rank=0):
     handles = []
     for name, p in params:
         if torch.is_tensor(p):
-            handles.append(comm.broadcast_async_(p.data, root_rank=root_rank,
+            t = p.data
+            if not t.is_contiguous():
+                t = utils.dense_flat_alias(t)
+            handles.append(comm.broadcast_async_(t, root_rank=root_rank,
                                                  name=name))
     for h in handles:
         comm.synchronize(h)

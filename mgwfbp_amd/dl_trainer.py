@@ -140,6 +140,13 @@ class DLTrainer:
                   'batch_size': self.batch_size}
         self.net, self.ext = create_net(self.num_classes, dnn, **kwargs)
         self.net = self.net.to(self.device)
+        # NHWC is MIOpen's fast conv path on CDNA4; grads keep their
+        # logical shapes so the bucket views are unaffected
+        self.channels_last = (
+            self.is_cuda and dnn not in ('lstm', 'lstman4', 'fcn5net', 'lr')
+            and os.environ.get('MGX_CHANNELS_LAST', '1') == '1')
+        if self.channels_last:
+            self.net = self.net.to(memory_format=torch.channels_last)
         if settings.DEBUG and rank == 0:
             logger.info('%s: %d parameters', dnn,
                         sum(p.numel() for p in self.net.parameters()))
@@ -296,7 +303,8 @@ class DLTrainer:
                      'mnist': (1, 28, 28)}[self.dataset]
             self._gpu_pool = mgx_data.GPUBatchPool.images(
                 self.batch_size, shape, self.num_classes, self.device,
-                n_batches=8, seed=self.rank)
+                n_batches=8, seed=self.rank,
+                channels_last=getattr(self, 'channels_last', False))
         if self._gpu_pool is not None:
             return self._gpu_pool.next()
         batch = self.data_iter()
@@ -306,6 +314,8 @@ class DLTrainer:
             return (inputs, input_sizes), (targets, target_sizes)
         inputs, labels = batch
         inputs = inputs.to(self.device, non_blocking=True)
+        if getattr(self, 'channels_last', False) and inputs.dim() == 4:
+            inputs = inputs.to(memory_format=torch.channels_last)
         labels = labels.to(self.device, non_blocking=True)
         return inputs, labels
 
