@@ -132,24 +132,13 @@ def main():
 
     elapsed = t1 - t0
     # MAX over ranks (slowest rank defines job time)
-    t = torch.tensor([elapsed], dtype=torch.float64)
     if comm.size() > 1:
-        if comm.backend_name() == 'rccl':
-            t = t.to(trainer.device)
-        h = comm.allreduce_async_(t, average=False)
-        comm.synchronize(h)
-        if torch.cuda.is_available():
-            torch.cuda.synchronize()
-        elapsed_sum = float(t.cpu()[0])
-        # max via allreduce of max: redo with explicit max reduction if
-        # backend supports; fall back to sum/size upper bound
         import torch.distributed as dist
-        if dist.is_initialized():
-            t2 = torch.tensor([elapsed], dtype=torch.float64)
-            dist.all_reduce(t2, op=dist.ReduceOp.MAX)
-            elapsed = float(t2[0])
-        else:
-            elapsed = elapsed_sum / comm.size()
+        t2 = torch.tensor([elapsed], dtype=torch.float64)
+        if dist.get_backend() == 'nccl':
+            t2 = t2.to(trainer.device)
+        dist.all_reduce(t2, op=dist.ReduceOp.MAX)
+        elapsed = float(t2.cpu()[0])
 
     ms_per_step = elapsed / args.steps * 1e3
     global_batch = args.batch_size * n_gpus

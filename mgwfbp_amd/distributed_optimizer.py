@@ -128,8 +128,10 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         nbytes = torch.tensor([], dtype=self._comm_dtype).element_size()
         a, b = solver.fit_alpha_beta([s * nbytes for s in sizes], times)
         t = torch.tensor([a, b], dtype=torch.float64)
+        if torch.cuda.is_available():
+            t = t.cuda()   # RCCL core broadcasts device tensors
         comm.broadcast(t, root_rank=0)
-        self.alpha, self.beta = float(t[0]), float(t[1])
+        self.alpha, self.beta = float(t[0].item()), float(t[1].item())
         logger.info('[rank %d] fitted allreduce model t = %.3e + %.3e * bytes',
                     comm.rank(), self.alpha, self.beta)
 
