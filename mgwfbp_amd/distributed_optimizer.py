@@ -184,9 +184,12 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         self._sizes = [self._named_parameters[k].numel()
                        for k in self._sequential_keys]
 
-        for gi, group in enumerate(groups):
-            new_key = ':'.join(group)       # reference key format (:285-288)
-            self._group_keys.append(new_key)
+        # ONE arena for all groups: zero_grad is a single memset and the
+        # per-group flat buffers are contiguous slices of it
+        group_offsets = []
+        group_totals = []
+        arena_total = 0
+        for group in groups:
             offsets = []
             off = 0
             for k in group:
@@ -194,9 +197,21 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                 n = self._named_parameters[k].numel()
                 off += (n + _GROUP_PAD - 1) // _GROUP_PAD * _GROUP_PAD
             total = off if off > 0 else _GROUP_PAD
-            first_p = self._named_parameters[group[0]]
-            flat = torch.zeros(total, dtype=torch.float32,
-                               device=first_p.device)
+            group_offsets.append(offsets)
+            group_totals.append(total)
+            arena_total += total
+        device = (self._named_parameters[self._sequential_keys[0]].device
+                  if self._sequential_keys else torch.device('cpu'))
+        self._grad_arena = torch.zeros(arena_total, dtype=torch.float32,
+                                       device=device)
+        arena_off = 0
+        for gi, group in enumerate(groups):
+            new_key = ':'.join(group)       # reference key format (:285-288)
+            self._group_keys.append(new_key)
+            offsets = group_offsets[gi]
+            total = group_totals[gi]
+            flat = self._grad_arena[arena_off:arena_off + total]
+            arena_off += total
             self._merged_parameters[new_key] = flat
             self._merged_parameter_names[flat] = new_key
             self._merged_parameter_offsets[new_key] = offsets
@@ -208,7 +223,7 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                 p.grad = utils.grad_view_like(flat[o:o + p.numel()], p)
             if self._comm_dtype != torch.float32:
                 self._group_comm_buffers[new_key] = torch.zeros(
-                    total, dtype=self._comm_dtype, device=first_p.device)
+                    total, dtype=self._comm_dtype, device=device)
                 if self._use_hip:
                     from . import kernels as _k
                     grads = [utils.dense_flat_alias(
@@ -357,10 +372,10 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         return super(self.__class__, self).step(closure)
 
     def zero_grad(self, set_to_none=False):
-        # grads are views into the flat buffers: zero the buffers (a few
-        # large memsets instead of one per tensor); never drop the views.
-        for flat in self._merged_parameters.values():
-            flat.zero_()
+        # grads are views into one arena: zero it with a SINGLE memset
+        # (the reference zeroes one tensor per parameter); never drop the
+        # views.
+        self._grad_arena.zero_()
 
     def load_state_dict(self, state_dict):
         super(self.__class__, self).load_state_dict(state_dict)
