@@ -155,6 +155,14 @@ class DLTrainer:
             and os.environ.get('MGX_CHANNELS_LAST', '1') == '1')
         if self.channels_last:
             self.net = self.net.to(memory_format=torch.channels_last)
+            if os.environ.get('MGX_FUSED_BN', '1') == '1':
+                try:
+                    from .kernels.batchnorm import convert_batchnorm
+                    from .kernels import hip_available
+                    if hip_available():
+                        convert_batchnorm(self.net)
+                except Exception as e:
+                    logger.warning('fused BN unavailable: %s', e)
         if settings.DEBUG and rank == 0:
             logger.info('%s: %d parameters', dnn,
                         sum(p.numel() for p in self.net.parameters()))

@@ -1,0 +1,489 @@
+// NHWC BatchNorm (training fwd/bwd + inference) for gfx950 / CDNA4.
+//
+// Why: rocprof on the ResNet-50 bf16 step (profiles/
+// resnet50_n1_bf16_kernel_stats.md) shows MIOpen's spatial BatchNorm --
+// 6 kernels per layer, run in fp32 with bf16<->fp32 cast kernels around
+// them (autocast promotes BN to fp32) -- costing 34% of the step. This
+// implementation consumes bf16/fp16 NHWC activations DIRECTLY with fp32
+// accumulation: ~2.5x fewer HBM bytes and no cast kernels.
+//
+// Layout contract: x is channels_last (N,H,W,C) dense, viewed as
+// rows = N*H*W by C. C must be a multiple of 8 (conv channels are).
+//
+// Thread mapping (all kernels): a block of 256 threads is a
+// (rows_per_block x octets_per_row) tile where octets_per_row = C/8
+// capped at 256: each active thread owns EIGHT consecutive channels
+// (one 16-byte bf16 octet -> coalesced 16 B/lane loads, guide G13) and
+// loops over rows; per-channel params live in registers for the whole
+// loop. Reduction: register accumulators -> LDS across the block's
+// rows -> one global fp32 atomicAdd per channel per block (guide G12).
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#define CHECK_HIP(x)                                                         \
+  do {                                                                       \
+    hipError_t err__ = (x);                                                  \
+    TORCH_CHECK(err__ == hipSuccess, "HIP error: ",                          \
+                hipGetErrorString(err__));                                   \
+  } while (0)
+
+namespace {
+
+constexpr int kBlock = 256;
+
+template <typename T> struct VecIO;
+template <> struct VecIO<float> {
+  // 8 floats = 32 B per lane
+  static __device__ inline void load(const float* p, float v[8]) {
+    const float4 a = *reinterpret_cast<const float4*>(p);
+    const float4 b = *reinterpret_cast<const float4*>(p + 4);
+    v[0] = a.x; v[1] = a.y; v[2] = a.z; v[3] = a.w;
+    v[4] = b.x; v[5] = b.y; v[6] = b.z; v[7] = b.w;
+  }
+  static __device__ inline void store(float* p, const float v[8]) {
+    *reinterpret_cast<float4*>(p) = {v[0], v[1], v[2], v[3]};
+    *reinterpret_cast<float4*>(p + 4) = {v[4], v[5], v[6], v[7]};
+  }
+};
+template <> struct VecIO<__hip_bfloat16> {
+  static __device__ inline void load(const __hip_bfloat16* p, float v[8]) {
+    const uint4 raw = *reinterpret_cast<const uint4*>(p);
+    const unsigned u[4] = {raw.x, raw.y, raw.z, raw.w};
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const __hip_bfloat162 h2 =
+          *reinterpret_cast<const __hip_bfloat162*>(&u[i]);
+      v[2 * i] = __bfloat162float(h2.x);
+      v[2 * i + 1] = __bfloat162float(h2.y);
+    }
+  }
+  static __device__ inline void store(__hip_bfloat16* p, const float v[8]) {
+    uint4 raw;
+    unsigned* u = reinterpret_cast<unsigned*>(&raw);
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      __hip_bfloat162 h2 = {__float2bfloat16(v[2 * i]),
+                            __float2bfloat16(v[2 * i + 1])};
+      u[i] = *reinterpret_cast<unsigned*>(&h2);
+    }
+    *reinterpret_cast<uint4*>(p) = raw;
+  }
+};
+template <> struct VecIO<__half> {
+  static __device__ inline void load(const __half* p, float v[8]) {
+    const uint4 raw = *reinterpret_cast<const uint4*>(p);
+    const unsigned u[4] = {raw.x, raw.y, raw.z, raw.w};
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const __half2 h2 = *reinterpret_cast<const __half2*>(&u[i]);
+      v[2 * i] = __half2float(h2.x);
+      v[2 * i + 1] = __half2float(h2.y);
+    }
+  }
+  static __device__ inline void store(__half* p, const float v[8]) {
+    uint4 raw;
+    unsigned* u = reinterpret_cast<unsigned*>(&raw);
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      __half2 h2 = __floats2half2_rn(v[2 * i], v[2 * i + 1]);
+      u[i] = *reinterpret_cast<unsigned*>(&h2);
+    }
+    *reinterpret_cast<uint4*>(p) = raw;
+  }
+};
+
+// octets_per_row = min(C/8, 256); rows_per_block = 256 / opr
+__device__ inline void tile_map(long C, int& opr, int& rpb, int& o,
+                                int& rl, bool& active) {
+  opr = (int)min(C / 8, (long)kBlock);
+  rpb = kBlock / opr;
+  o = threadIdx.x % opr;
+  rl = threadIdx.x / opr;
+  active = (rl < rpb);
+}
+
+// -------------------------------------------------- fwd: reduce ---------
+// partial[2C]: sum in [0,C), sumsq in [C,2C) -- zeroed before launch.
+template <typename T>
+__global__ __launch_bounds__(kBlock) void bn_fwd_reduce_kernel(
+    const T* __restrict__ x, long rows, long C,
+    float* __restrict__ partial) {
+  int opr, rpb, o, rl; bool active;
+  tile_map(C, opr, rpb, o, rl, active);
+  const long octets = C / 8;
+  float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float q[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (active) {
+    for (long ob = o; ob < octets; ob += opr) {     // C > 2048 spillover
+      const long c0 = ob * 8;
+      for (long r = (long)blockIdx.x * rpb + rl; r < rows;
+           r += (long)gridDim.x * rpb) {
+        float v[8];
+        VecIO<T>::load(x + r * C + c0, v);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          s[i] += v[i];
+          q[i] += v[i] * v[i];
+        }
+      }
+      if (octets > opr) {   // rare wide-C path: flush per octet
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          atomicAdd(&partial[c0 + i], s[i]);
+          atomicAdd(&partial[C + c0 + i], q[i]);
+          s[i] = 0; q[i] = 0;
+        }
+      }
+    }
+  }
+  if (C / 8 > kBlock) return;      // wide-C path already flushed
+  // block reduction: stash partials in LDS, row 0 of each octet sums
+  __shared__ float lds[kBlock * 16];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    lds[(rl * opr + o) * 16 + i] = active ? s[i] : 0.f;
+    lds[(rl * opr + o) * 16 + 8 + i] = active ? q[i] : 0.f;
+  }
+  __syncthreads();
+  if (rl == 0 && active) {
+    float fs[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    float fq[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int r = 0; r < rpb; ++r) {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        fs[i] += lds[(r * opr + o) * 16 + i];
+        fq[i] += lds[(r * opr + o) * 16 + 8 + i];
+      }
+    }
+    const long c0 = (long)o * 8;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      atomicAdd(&partial[c0 + i], fs[i]);
+      atomicAdd(&partial[C + c0 + i], fq[i]);
+    }
+  }
+}
+
+// ------------------------------------------------ fwd: finalize ---------
+// mean/invstd from partials + running-stat update (one tiny launch).
+__global__ __launch_bounds__(kBlock) void bn_fwd_finalize_kernel(
+    const float* __restrict__ partial, long C, float M, float eps,
+    float momentum, float* __restrict__ mean, float* __restrict__ invstd,
+    float* __restrict__ running_mean, float* __restrict__ running_var) {
+  for (long c = blockIdx.x * blockDim.x + threadIdx.x; c < C;
+       c += (long)gridDim.x * blockDim.x) {
+    const float m = partial[c] / M;
+    float var = partial[C + c] / M - m * m;
+    var = var < 0.f ? 0.f : var;
+    mean[c] = m;
+    invstd[c] = rsqrtf(var + eps);
+    if (running_mean != nullptr) {
+      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+      const float unbiased = M > 1.f ? var * M / (M - 1.f) : var;
+      running_var[c] =
+          (1.f - momentum) * running_var[c] + momentum * unbiased;
+    }
+  }
+}
+
+// --------------------------------------------------- fwd: norm ----------
+template <typename T>
+__global__ __launch_bounds__(kBlock) void bn_fwd_norm_kernel(
+    const T* __restrict__ x, T* __restrict__ y, long rows, long C,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    bool relu) {
+  int opr, rpb, o, rl; bool active;
+  tile_map(C, opr, rpb, o, rl, active);
+  if (!active) return;
+  const long octets = C / 8;
+  for (long ob = o; ob < octets; ob += opr) {
+    const long c0 = ob * 8;
+    float sc[8], sh[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      const float g = gamma ? gamma[c0 + i] : 1.f;
+      sc[i] = g * invstd[c0 + i];
+      sh[i] = (beta ? beta[c0 + i] : 0.f) - mean[c0 + i] * sc[i];
+    }
+    for (long r = (long)blockIdx.x * rpb + rl; r < rows;
+         r += (long)gridDim.x * rpb) {
+      float v[8];
+      VecIO<T>::load(x + r * C + c0, v);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        v[i] = v[i] * sc[i] + sh[i];
+        if (relu) v[i] = v[i] > 0.f ? v[i] : 0.f;
+      }
+      VecIO<T>::store(y + r * C + c0, v);
+    }
+  }
+}
+
+// -------------------------------------------------- bwd: reduce ---------
+// partial[2C]: sum(dy) in [0,C), sum(dy*xhat) in [C,2C).
+// relu: dy is gated by y>0, recomputed from (x, mean, invstd, gamma,
+// beta) so no extra saved tensor is needed.
+template <typename T>
+__global__ __launch_bounds__(kBlock) void bn_bwd_reduce_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x, long rows, long C,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    float* __restrict__ partial) {
+  int opr, rpb, o, rl; bool active;
+  tile_map(C, opr, rpb, o, rl, active);
+  const long octets = C / 8;
+  float sd[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float sx[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (active) {
+    for (long ob = o; ob < octets; ob += opr) {
+      const long c0 = ob * 8;
+      float mu[8], is[8];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        mu[i] = mean[c0 + i];
+        is[i] = invstd[c0 + i];
+      }
+      for (long r = (long)blockIdx.x * rpb + rl; r < rows;
+           r += (long)gridDim.x * rpb) {
+        float g[8], v[8];
+        VecIO<T>::load(dy + r * C + c0, g);
+        VecIO<T>::load(x + r * C + c0, v);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          const float xh = (v[i] - mu[i]) * is[i];
+          sd[i] += g[i];
+          sx[i] += g[i] * xh;
+        }
+      }
+      if (octets > opr) {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          atomicAdd(&partial[c0 + i], sd[i]);
+          atomicAdd(&partial[C + c0 + i], sx[i]);
+          sd[i] = 0; sx[i] = 0;
+        }
+      }
+    }
+  }
+  if (C / 8 > kBlock) return;
+  __shared__ float lds[kBlock * 16];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    lds[(rl * opr + o) * 16 + i] = active ? sd[i] : 0.f;
+    lds[(rl * opr + o) * 16 + 8 + i] = active ? sx[i] : 0.f;
+  }
+  __syncthreads();
+  if (rl == 0 && active) {
+    float fd[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    float fx[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int r = 0; r < rpb; ++r) {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        fd[i] += lds[(r * opr + o) * 16 + i];
+        fx[i] += lds[(r * opr + o) * 16 + 8 + i];
+      }
+    }
+    const long c0 = (long)o * 8;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      atomicAdd(&partial[c0 + i], fd[i]);
+      atomicAdd(&partial[C + c0 + i], fx[i]);
+    }
+  }
+}
+
+// ---------------------------------------------------- bwd: dx -----------
+// dx = gamma*invstd*(dy - sum_dy/M - xhat*sum_dy_xhat/M)
+template <typename T>
+__global__ __launch_bounds__(kBlock) void bn_bwd_dx_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x, T* __restrict__ dx,
+    long rows, long C, const float* __restrict__ mean,
+    const float* __restrict__ invstd, const float* __restrict__ gamma,
+    const float* __restrict__ partial, float invM) {
+  int opr, rpb, o, rl; bool active;
+  tile_map(C, opr, rpb, o, rl, active);
+  if (!active) return;
+  const long octets = C / 8;
+  for (long ob = o; ob < octets; ob += opr) {
+    const long c0 = ob * 8;
+    float mu[8], is[8], gi[8], md[8], mx[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      mu[i] = mean[c0 + i];
+      is[i] = invstd[c0 + i];
+      gi[i] = (gamma ? gamma[c0 + i] : 1.f) * is[i];
+      md[i] = partial[c0 + i] * invM;          // mean of dy
+      mx[i] = partial[C + c0 + i] * invM;      // mean of dy*xhat
+    }
+    for (long r = (long)blockIdx.x * rpb + rl; r < rows;
+         r += (long)gridDim.x * rpb) {
+      float g[8], v[8];
+      VecIO<T>::load(dy + r * C + c0, g);
+      VecIO<T>::load(x + r * C + c0, v);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const float xh = (v[i] - mu[i]) * is[i];
+        g[i] = gi[i] * (g[i] - md[i] - xh * mx[i]);
+      }
+      VecIO<T>::store(dx + r * C + c0, g);
+    }
+  }
+}
+
+// ------------------------------------------------- host plumbing --------
+
+int grid_rows(long rows, int rpb, long C) {
+  // enough blocks to fill 256 CUs, few enough that the per-block atomic
+  // tail (2C atomics) stays cheap for small C
+  long nblk = (rows + rpb - 1) / rpb;
+  long cap = C >= 256 ? 1536 : 640;
+  if (nblk > cap) nblk = cap;
+  return (int)(nblk < 1 ? 1 : nblk);
+}
+
+struct Geometry {
+  long rows, C;
+  int opr, rpb;
+};
+
+Geometry geom(const torch::Tensor& x) {
+  TORCH_CHECK(x.dim() == 4, "expected 4-D NHWC tensor");
+  const long N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  TORCH_CHECK(C % 8 == 0, "C must be a multiple of 8");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "expected channels_last input");
+  Geometry g;
+  g.rows = N * H * W;
+  g.C = C;
+  g.opr = (int)std::min(C / 8, (long)kBlock);
+  g.rpb = kBlock / g.opr;
+  return g;
+}
+
+#define DISPATCH_DT(scalar_type, ...)                                        \
+  switch (scalar_type) {                                                     \
+    case at::kFloat: {                                                       \
+      using dt = float; __VA_ARGS__; break;                                  \
+    }                                                                        \
+    case at::kBFloat16: {                                                    \
+      using dt = __hip_bfloat16; __VA_ARGS__; break;                         \
+    }                                                                        \
+    case at::kHalf: {                                                        \
+      using dt = __half; __VA_ARGS__; break;                                 \
+    }                                                                        \
+    default: TORCH_CHECK(false, "unsupported dtype");                        \
+  }
+
+std::vector<torch::Tensor> bn_fwd_train(
+    torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
+    torch::Tensor running_mean, torch::Tensor running_var, double momentum,
+    double eps, bool relu) {
+  auto g = geom(x);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  auto opts = torch::TensorOptions().dtype(torch::kFloat).device(x.device());
+  auto partial = torch::zeros({2 * g.C}, opts);
+  auto mean = torch::empty({g.C}, opts);
+  auto invstd = torch::empty({g.C}, opts);
+  auto y = torch::empty_like(x);
+  const int blocks = grid_rows(g.rows, g.rpb, g.C);
+  DISPATCH_DT(x.scalar_type(), {
+    hipLaunchKernelGGL(bn_fwd_reduce_kernel<dt>, dim3(blocks), dim3(kBlock),
+                       0, stream,
+                       reinterpret_cast<const dt*>(x.data_ptr()), g.rows,
+                       g.C, partial.data_ptr<float>());
+  });
+  hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3(1), dim3(kBlock), 0,
+                     stream, partial.data_ptr<float>(), g.C,
+                     (float)g.rows, (float)eps, (float)momentum,
+                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                     running_mean.defined()
+                         ? running_mean.data_ptr<float>() : nullptr,
+                     running_var.defined()
+                         ? running_var.data_ptr<float>() : nullptr);
+  DISPATCH_DT(x.scalar_type(), {
+    hipLaunchKernelGGL(bn_fwd_norm_kernel<dt>, dim3(blocks), dim3(kBlock),
+                       0, stream,
+                       reinterpret_cast<const dt*>(x.data_ptr()),
+                       reinterpret_cast<dt*>(y.data_ptr()), g.rows, g.C,
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       gamma.defined() ? gamma.data_ptr<float>() : nullptr,
+                       beta.defined() ? beta.data_ptr<float>() : nullptr,
+                       relu);
+  });
+  CHECK_HIP(hipGetLastError());
+  return {y, mean, invstd};
+}
+
+torch::Tensor bn_fwd_eval(torch::Tensor x, torch::Tensor gamma,
+                          torch::Tensor beta, torch::Tensor running_mean,
+                          torch::Tensor running_var, double eps,
+                          bool relu) {
+  auto g = geom(x);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  auto invstd = (running_var + eps).rsqrt();
+  auto y = torch::empty_like(x);
+  const int blocks = grid_rows(g.rows, g.rpb, g.C);
+  DISPATCH_DT(x.scalar_type(), {
+    hipLaunchKernelGGL(bn_fwd_norm_kernel<dt>, dim3(blocks), dim3(kBlock),
+                       0, stream,
+                       reinterpret_cast<const dt*>(x.data_ptr()),
+                       reinterpret_cast<dt*>(y.data_ptr()), g.rows, g.C,
+                       running_mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(),
+                       gamma.defined() ? gamma.data_ptr<float>() : nullptr,
+                       beta.defined() ? beta.data_ptr<float>() : nullptr,
+                       relu);
+  });
+  CHECK_HIP(hipGetLastError());
+  return y;
+}
+
+std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
+                                  torch::Tensor mean, torch::Tensor invstd,
+                                  torch::Tensor gamma) {
+  auto g = geom(x);
+  TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "expected channels_last grad");
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  auto opts = torch::TensorOptions().dtype(torch::kFloat).device(x.device());
+  auto partial = torch::zeros({2 * g.C}, opts);
+  auto dx = torch::empty_like(x);
+  const int blocks = grid_rows(g.rows, g.rpb, g.C);
+  DISPATCH_DT(x.scalar_type(), {
+    hipLaunchKernelGGL(bn_bwd_reduce_kernel<dt>, dim3(blocks), dim3(kBlock),
+                       0, stream,
+                       reinterpret_cast<const dt*>(dy.data_ptr()),
+                       reinterpret_cast<const dt*>(x.data_ptr()), g.rows,
+                       g.C, mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(),
+                       partial.data_ptr<float>());
+  });
+  DISPATCH_DT(x.scalar_type(), {
+    hipLaunchKernelGGL(bn_bwd_dx_kernel<dt>, dim3(blocks), dim3(kBlock), 0,
+                       stream,
+                       reinterpret_cast<const dt*>(dy.data_ptr()),
+                       reinterpret_cast<const dt*>(x.data_ptr()),
+                       reinterpret_cast<dt*>(dx.data_ptr()), g.rows, g.C,
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       gamma.defined() ? gamma.data_ptr<float>() : nullptr,
+                       partial.data_ptr<float>(), 1.f / (float)g.rows);
+  });
+  CHECK_HIP(hipGetLastError());
+  // dgamma = partial[C:2C], dbeta = partial[0:C]
+  auto dbeta = partial.narrow(0, 0, g.C).clone();
+  auto dgamma = partial.narrow(0, g.C, g.C).clone();
+  return {dx, dgamma, dbeta};
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("bn_fwd_train", &bn_fwd_train,
+        "NHWC BN training forward; returns (y, mean, invstd)");
+  m.def("bn_fwd_eval", &bn_fwd_eval, "NHWC BN inference forward");
+  m.def("bn_bwd", &bn_bwd,
+        "NHWC BN backward; returns (dx, dgamma, dbeta)");
+}

@@ -31,6 +31,11 @@ ext_modules = [
         extra_compile_args=common_args,
     ),
     CUDAExtension(
+        name='mgwfbp_amd.kernels.mgx_bn_ext',
+        sources=['mgwfbp_amd/kernels/bn_kernels.hip'],
+        extra_compile_args=common_args,
+    ),
+    CUDAExtension(
         name='mgwfbp_amd.comm.mgx_comm_ext',
         sources=['mgwfbp_amd/comm/comm_core.hip'],
         libraries=['rccl'],
