@@ -43,6 +43,38 @@ def time_one(mod, x, iters=20):
     return tf / iters, tb / iters
 
 
+def time_raw(x, iters=20):
+    """Raw extension calls without autograd, to isolate host overhead."""
+    from mgwfbp_amd.kernels.batchnorm import _load
+    ext = _load()
+    C = x.size(1)
+    w = torch.ones(C, device='cuda')
+    b = torch.zeros(C, device='cuda')
+    rm = torch.zeros(C, device='cuda')
+    rv = torch.ones(C, device='cuda')
+    dy = torch.randn_like(x)
+    for _ in range(3):
+        y, mean, invstd = ext.bn_fwd_train(x, w, b, rm, rv, 0.1, 1e-5,
+                                           False)
+        ext.bn_bwd(dy, x, mean, invstd, w)
+    torch.cuda.synchronize()
+    s = torch.cuda.Event(True)
+    m = torch.cuda.Event(True)
+    e = torch.cuda.Event(True)
+    tf = tb = 0.0
+    for _ in range(iters):
+        s.record()
+        y, mean, invstd = ext.bn_fwd_train(x, w, b, rm, rv, 0.1, 1e-5,
+                                           False)
+        m.record()
+        ext.bn_bwd(dy, x, mean, invstd, w)
+        e.record()
+        torch.cuda.synchronize()
+        tf += s.elapsed_time(m)
+        tb += m.elapsed_time(e)
+    return tf / iters, tb / iters
+
+
 def main():
     dtype = torch.bfloat16 if len(sys.argv) < 2 else \
         {'fp32': torch.float32, 'bf16': torch.bfloat16}[sys.argv[1]]
@@ -59,8 +91,9 @@ def main():
                             enabled=dtype == torch.bfloat16):
             rf, rb = time_one(ref, x)
             of, ob = time_one(ours, x)
-        print('%22s %9.3f  %9.3f  %9.3f  %9.3f ms'
-              % (str((C, H, W)), rf, rb, of, ob))
+        xf, xb = time_raw(x.detach())
+        print('%22s %9.3f %9.3f %9.3f %9.3f | raw %6.3f %6.3f ms'
+              % (str((C, H, W)), rf, rb, of, ob, xf, xb))
         tot_ref += rf + rb
         tot_ours += of + ob
     print('TOTAL per-pass: ref %.3f ms  mgx %.3f ms' % (tot_ref, tot_ours))
