@@ -8,15 +8,21 @@
 // accumulation: ~2.5x fewer HBM bytes and no cast kernels.
 //
 // Layout contract: x is channels_last (N,H,W,C) dense, viewed as
-// rows = N*H*W by C. C must be a multiple of 8 (conv channels are).
+// rows = N*H*W by C. C must be a multiple of 8.
 //
 // Thread mapping (all kernels): a block of 256 threads is a
 // (rows_per_block x octets_per_row) tile where octets_per_row = C/8
 // capped at 256: each active thread owns EIGHT consecutive channels
 // (one 16-byte bf16 octet -> coalesced 16 B/lane loads, guide G13) and
-// loops over rows; per-channel params live in registers for the whole
-// loop. Reduction: register accumulators -> LDS across the block's
-// rows -> one global fp32 atomicAdd per channel per block (guide G12).
+// loops over rows with per-channel params held in registers.
+//
+// Reduction is TWO-STAGE with a per-block partials buffer -- register
+// accumulators -> LDS across the block's rows -> ONE plain store of the
+// block's 2C partial sums -> a finalize kernel sums over blocks. No
+// global atomics anywhere: a first version used one atomicAdd per
+// channel per block and the same-address serialization made small
+// layers ~30x slower than the data movement (guide G12's "reduce then
+// one atomic" still serializes when blocks >> channels).
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 #include <hip/hip_fp16.h>
@@ -36,7 +42,6 @@ constexpr int kBlock = 256;
 
 template <typename T> struct VecIO;
 template <> struct VecIO<float> {
-  // 8 floats = 32 B per lane
   static __device__ inline void load(const float* p, float v[8]) {
     const float4 a = *reinterpret_cast<const float4*>(p);
     const float4 b = *reinterpret_cast<const float4*>(p + 4);
@@ -95,7 +100,6 @@ template <> struct VecIO<__half> {
   }
 };
 
-// octets_per_row = min(C/8, 256); rows_per_block = 256 / opr
 __device__ inline void tile_map(long C, int& opr, int& rpb, int& o,
                                 int& rl, bool& active) {
   opr = (int)min(C / 8, (long)kBlock);
@@ -105,8 +109,53 @@ __device__ inline void tile_map(long C, int& opr, int& rpb, int& o,
   active = (rl < rpb);
 }
 
+// Generic per-block reduction of two 8-wide accumulators (a, b) keyed by
+// channel octet, then ONE plain store per channel into
+// partial[blockIdx.x * 2C .. +2C) (sum_a in [0,C), sum_b in [C,2C)).
+__device__ inline void block_reduce_store(float a[8], float b[8], long C,
+                                          int opr, int rpb, int o, int rl,
+                                          bool active, long c0_wide,
+                                          bool wide,
+                                          float* __restrict__ partial) {
+  float* out = partial + (long)blockIdx.x * 2 * C;
+  if (wide) {
+    // C > 2048: each octet owned by exactly one thread in the block
+    if (active) {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        out[c0_wide + i] = a[i];
+        out[C + c0_wide + i] = b[i];
+      }
+    }
+    return;
+  }
+  __shared__ float lds[kBlock * 16];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    lds[(rl * opr + o) * 16 + i] = active ? a[i] : 0.f;
+    lds[(rl * opr + o) * 16 + 8 + i] = active ? b[i] : 0.f;
+  }
+  __syncthreads();
+  if (rl == 0 && active) {
+    float fa[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    float fb[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int r = 0; r < rpb; ++r) {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        fa[i] += lds[(r * opr + o) * 16 + i];
+        fb[i] += lds[(r * opr + o) * 16 + 8 + i];
+      }
+    }
+    const long c0 = (long)o * 8;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      out[c0 + i] = fa[i];
+      out[C + c0 + i] = fb[i];
+    }
+  }
+}
+
 // -------------------------------------------------- fwd: reduce ---------
-// partial[2C]: sum in [0,C), sumsq in [C,2C) -- zeroed before launch.
 template <typename T>
 __global__ __launch_bounds__(kBlock) void bn_fwd_reduce_kernel(
     const T* __restrict__ x, long rows, long C,
@@ -114,11 +163,14 @@ __global__ __launch_bounds__(kBlock) void bn_fwd_reduce_kernel(
   int opr, rpb, o, rl; bool active;
   tile_map(C, opr, rpb, o, rl, active);
   const long octets = C / 8;
+  const bool wide = octets > opr;
   float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   float q[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  long c0_last = 0;
   if (active) {
-    for (long ob = o; ob < octets; ob += opr) {     // C > 2048 spillover
+    for (long ob = o; ob < octets; ob += opr) {
       const long c0 = ob * 8;
+      c0_last = c0;
       for (long r = (long)blockIdx.x * rpb + rl; r < rows;
            r += (long)gridDim.x * rpb) {
         float v[8];
@@ -129,54 +181,38 @@ __global__ __launch_bounds__(kBlock) void bn_fwd_reduce_kernel(
           q[i] += v[i] * v[i];
         }
       }
-      if (octets > opr) {   // rare wide-C path: flush per octet
+      if (wide) {
+        float* out = partial + (long)blockIdx.x * 2 * C;
 #pragma unroll
         for (int i = 0; i < 8; ++i) {
-          atomicAdd(&partial[c0 + i], s[i]);
-          atomicAdd(&partial[C + c0 + i], q[i]);
+          out[c0 + i] = s[i];
+          out[C + c0 + i] = q[i];
           s[i] = 0; q[i] = 0;
         }
       }
     }
   }
-  if (C / 8 > kBlock) return;      // wide-C path already flushed
-  // block reduction: stash partials in LDS, row 0 of each octet sums
-  __shared__ float lds[kBlock * 16];
-#pragma unroll
-  for (int i = 0; i < 8; ++i) {
-    lds[(rl * opr + o) * 16 + i] = active ? s[i] : 0.f;
-    lds[(rl * opr + o) * 16 + 8 + i] = active ? q[i] : 0.f;
-  }
-  __syncthreads();
-  if (rl == 0 && active) {
-    float fs[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    float fq[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    for (int r = 0; r < rpb; ++r) {
-#pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        fs[i] += lds[(r * opr + o) * 16 + i];
-        fq[i] += lds[(r * opr + o) * 16 + 8 + i];
-      }
-    }
-    const long c0 = (long)o * 8;
-#pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      atomicAdd(&partial[c0 + i], fs[i]);
-      atomicAdd(&partial[C + c0 + i], fq[i]);
-    }
-  }
+  if (!wide)
+    block_reduce_store(s, q, C, opr, rpb, o, rl, active, c0_last, false,
+                       partial);
 }
 
 // ------------------------------------------------ fwd: finalize ---------
-// mean/invstd from partials + running-stat update (one tiny launch).
+// Sum the per-block partials, then mean/invstd + running stats.
 __global__ __launch_bounds__(kBlock) void bn_fwd_finalize_kernel(
-    const float* __restrict__ partial, long C, float M, float eps,
-    float momentum, float* __restrict__ mean, float* __restrict__ invstd,
-    float* __restrict__ running_mean, float* __restrict__ running_var) {
+    const float* __restrict__ partial, long nblk, long C, float M,
+    float eps, float momentum, float* __restrict__ mean,
+    float* __restrict__ invstd, float* __restrict__ running_mean,
+    float* __restrict__ running_var) {
   for (long c = blockIdx.x * blockDim.x + threadIdx.x; c < C;
        c += (long)gridDim.x * blockDim.x) {
-    const float m = partial[c] / M;
-    float var = partial[C + c] / M - m * m;
+    float s = 0.f, q = 0.f;
+    for (long b = 0; b < nblk; ++b) {
+      s += partial[b * 2 * C + c];
+      q += partial[b * 2 * C + C + c];
+    }
+    const float m = s / M;
+    float var = q / M - m * m;
     var = var < 0.f ? 0.f : var;
     mean[c] = m;
     invstd[c] = rsqrtf(var + eps);
@@ -224,9 +260,6 @@ __global__ __launch_bounds__(kBlock) void bn_fwd_norm_kernel(
 }
 
 // -------------------------------------------------- bwd: reduce ---------
-// partial[2C]: sum(dy) in [0,C), sum(dy*xhat) in [C,2C).
-// relu: dy is gated by y>0, recomputed from (x, mean, invstd, gamma,
-// beta) so no extra saved tensor is needed.
 template <typename T>
 __global__ __launch_bounds__(kBlock) void bn_bwd_reduce_kernel(
     const T* __restrict__ dy, const T* __restrict__ x, long rows, long C,
@@ -235,11 +268,14 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_reduce_kernel(
   int opr, rpb, o, rl; bool active;
   tile_map(C, opr, rpb, o, rl, active);
   const long octets = C / 8;
+  const bool wide = octets > opr;
   float sd[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   float sx[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  long c0_last = 0;
   if (active) {
     for (long ob = o; ob < octets; ob += opr) {
       const long c0 = ob * 8;
+      c0_last = c0;
       float mu[8], is[8];
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
@@ -258,51 +294,47 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_reduce_kernel(
           sx[i] += g[i] * xh;
         }
       }
-      if (octets > opr) {
+      if (wide) {
+        float* out = partial + (long)blockIdx.x * 2 * C;
 #pragma unroll
         for (int i = 0; i < 8; ++i) {
-          atomicAdd(&partial[c0 + i], sd[i]);
-          atomicAdd(&partial[C + c0 + i], sx[i]);
+          out[c0 + i] = sd[i];
+          out[C + c0 + i] = sx[i];
           sd[i] = 0; sx[i] = 0;
         }
       }
     }
   }
-  if (C / 8 > kBlock) return;
-  __shared__ float lds[kBlock * 16];
-#pragma unroll
-  for (int i = 0; i < 8; ++i) {
-    lds[(rl * opr + o) * 16 + i] = active ? sd[i] : 0.f;
-    lds[(rl * opr + o) * 16 + 8 + i] = active ? sx[i] : 0.f;
-  }
-  __syncthreads();
-  if (rl == 0 && active) {
-    float fd[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    float fx[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    for (int r = 0; r < rpb; ++r) {
-#pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        fd[i] += lds[(r * opr + o) * 16 + i];
-        fx[i] += lds[(r * opr + o) * 16 + 8 + i];
-      }
+  if (!wide)
+    block_reduce_store(sd, sx, C, opr, rpb, o, rl, active, c0_last, false,
+                       partial);
+}
+
+// --------------------------------------------- bwd: finalize ------------
+__global__ __launch_bounds__(kBlock) void bn_bwd_finalize_kernel(
+    const float* __restrict__ partial, long nblk, long C,
+    float* __restrict__ dbeta, float* __restrict__ dgamma) {
+  for (long c = blockIdx.x * blockDim.x + threadIdx.x; c < C;
+       c += (long)gridDim.x * blockDim.x) {
+    float sd = 0.f, sx = 0.f;
+    for (long b = 0; b < nblk; ++b) {
+      sd += partial[b * 2 * C + c];
+      sx += partial[b * 2 * C + C + c];
     }
-    const long c0 = (long)o * 8;
-#pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      atomicAdd(&partial[c0 + i], fd[i]);
-      atomicAdd(&partial[C + c0 + i], fx[i]);
-    }
+    dbeta[c] = sd;
+    dgamma[c] = sx;
   }
 }
 
 // ---------------------------------------------------- bwd: dx -----------
-// dx = gamma*invstd*(dy - sum_dy/M - xhat*sum_dy_xhat/M)
+// dx = gamma*invstd*(dy - dbeta/M - xhat*dgamma/M)
 template <typename T>
 __global__ __launch_bounds__(kBlock) void bn_bwd_dx_kernel(
     const T* __restrict__ dy, const T* __restrict__ x, T* __restrict__ dx,
     long rows, long C, const float* __restrict__ mean,
     const float* __restrict__ invstd, const float* __restrict__ gamma,
-    const float* __restrict__ partial, float invM) {
+    const float* __restrict__ dbeta, const float* __restrict__ dgamma,
+    float invM) {
   int opr, rpb, o, rl; bool active;
   tile_map(C, opr, rpb, o, rl, active);
   if (!active) return;
@@ -315,8 +347,8 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_dx_kernel(
       mu[i] = mean[c0 + i];
       is[i] = invstd[c0 + i];
       gi[i] = (gamma ? gamma[c0 + i] : 1.f) * is[i];
-      md[i] = partial[c0 + i] * invM;          // mean of dy
-      mx[i] = partial[C + c0 + i] * invM;      // mean of dy*xhat
+      md[i] = dbeta[c0 + i] * invM;
+      mx[i] = dgamma[c0 + i] * invM;
     }
     for (long r = (long)blockIdx.x * rpb + rl; r < rows;
          r += (long)gridDim.x * rpb) {
@@ -335,15 +367,6 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_dx_kernel(
 
 // ------------------------------------------------- host plumbing --------
 
-int grid_rows(long rows, int rpb, long C) {
-  // enough blocks to fill 256 CUs, few enough that the per-block atomic
-  // tail (2C atomics) stays cheap for small C
-  long nblk = (rows + rpb - 1) / rpb;
-  long cap = C >= 256 ? 1536 : 640;
-  if (nblk > cap) nblk = cap;
-  return (int)(nblk < 1 ? 1 : nblk);
-}
-
 struct Geometry {
   long rows, C;
   int opr, rpb;
@@ -361,6 +384,22 @@ Geometry geom(const torch::Tensor& x) {
   g.opr = (int)std::min(C / 8, (long)kBlock);
   g.rpb = kBlock / g.opr;
   return g;
+}
+
+// Reduce-kernel grid: stream-filling for big inputs, but never more
+// blocks than ~rows/(rpb*8) so the finalize pass over [nblk][2C]
+// partials stays negligible.
+int grid_reduce(const Geometry& g) {
+  long nblk = (g.rows + (long)g.rpb * 8 - 1) / ((long)g.rpb * 8);
+  if (nblk > 1024) nblk = 1024;
+  return (int)(nblk < 1 ? 1 : nblk);
+}
+
+// Elementwise-kernel grid: pure streaming, fill the chip.
+int grid_elem(const Geometry& g) {
+  long nblk = (g.rows + g.rpb - 1) / g.rpb;
+  if (nblk > 2048) nblk = 2048;
+  return (int)(nblk < 1 ? 1 : nblk);
 }
 
 #define DISPATCH_DT(scalar_type, ...)                                        \
@@ -384,19 +423,21 @@ std::vector<torch::Tensor> bn_fwd_train(
   auto g = geom(x);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   auto opts = torch::TensorOptions().dtype(torch::kFloat).device(x.device());
-  auto partial = torch::zeros({2 * g.C}, opts);
+  const int rblocks = grid_reduce(g);
+  const int eblocks = grid_elem(g);
+  auto partial = torch::empty({(long)rblocks * 2 * g.C}, opts);
   auto mean = torch::empty({g.C}, opts);
   auto invstd = torch::empty({g.C}, opts);
   auto y = torch::empty_like(x);
-  const int blocks = grid_rows(g.rows, g.rpb, g.C);
   DISPATCH_DT(x.scalar_type(), {
-    hipLaunchKernelGGL(bn_fwd_reduce_kernel<dt>, dim3(blocks), dim3(kBlock),
-                       0, stream,
+    hipLaunchKernelGGL(bn_fwd_reduce_kernel<dt>, dim3(rblocks),
+                       dim3(kBlock), 0, stream,
                        reinterpret_cast<const dt*>(x.data_ptr()), g.rows,
                        g.C, partial.data_ptr<float>());
   });
-  hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3(1), dim3(kBlock), 0,
-                     stream, partial.data_ptr<float>(), g.C,
+  hipLaunchKernelGGL(bn_fwd_finalize_kernel,
+                     dim3((g.C + kBlock - 1) / kBlock), dim3(kBlock), 0,
+                     stream, partial.data_ptr<float>(), rblocks, g.C,
                      (float)g.rows, (float)eps, (float)momentum,
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      running_mean.defined()
@@ -404,7 +445,7 @@ std::vector<torch::Tensor> bn_fwd_train(
                      running_var.defined()
                          ? running_var.data_ptr<float>() : nullptr);
   DISPATCH_DT(x.scalar_type(), {
-    hipLaunchKernelGGL(bn_fwd_norm_kernel<dt>, dim3(blocks), dim3(kBlock),
+    hipLaunchKernelGGL(bn_fwd_norm_kernel<dt>, dim3(eblocks), dim3(kBlock),
                        0, stream,
                        reinterpret_cast<const dt*>(x.data_ptr()),
                        reinterpret_cast<dt*>(y.data_ptr()), g.rows, g.C,
@@ -425,9 +466,9 @@ torch::Tensor bn_fwd_eval(torch::Tensor x, torch::Tensor gamma,
   auto stream = c10::hip::getCurrentHIPStream().stream();
   auto invstd = (running_var + eps).rsqrt();
   auto y = torch::empty_like(x);
-  const int blocks = grid_rows(g.rows, g.rpb, g.C);
+  const int eblocks = grid_elem(g);
   DISPATCH_DT(x.scalar_type(), {
-    hipLaunchKernelGGL(bn_fwd_norm_kernel<dt>, dim3(blocks), dim3(kBlock),
+    hipLaunchKernelGGL(bn_fwd_norm_kernel<dt>, dim3(eblocks), dim3(kBlock),
                        0, stream,
                        reinterpret_cast<const dt*>(x.data_ptr()),
                        reinterpret_cast<dt*>(y.data_ptr()), g.rows, g.C,
@@ -449,32 +490,37 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
               "expected channels_last grad");
   auto stream = c10::hip::getCurrentHIPStream().stream();
   auto opts = torch::TensorOptions().dtype(torch::kFloat).device(x.device());
-  auto partial = torch::zeros({2 * g.C}, opts);
+  const int rblocks = grid_reduce(g);
+  const int eblocks = grid_elem(g);
+  auto partial = torch::empty({(long)rblocks * 2 * g.C}, opts);
+  auto dbeta = torch::empty({g.C}, opts);
+  auto dgamma = torch::empty({g.C}, opts);
   auto dx = torch::empty_like(x);
-  const int blocks = grid_rows(g.rows, g.rpb, g.C);
   DISPATCH_DT(x.scalar_type(), {
-    hipLaunchKernelGGL(bn_bwd_reduce_kernel<dt>, dim3(blocks), dim3(kBlock),
-                       0, stream,
+    hipLaunchKernelGGL(bn_bwd_reduce_kernel<dt>, dim3(rblocks),
+                       dim3(kBlock), 0, stream,
                        reinterpret_cast<const dt*>(dy.data_ptr()),
                        reinterpret_cast<const dt*>(x.data_ptr()), g.rows,
                        g.C, mean.data_ptr<float>(),
                        invstd.data_ptr<float>(),
                        partial.data_ptr<float>());
   });
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel,
+                     dim3((g.C + kBlock - 1) / kBlock), dim3(kBlock), 0,
+                     stream, partial.data_ptr<float>(), rblocks, g.C,
+                     dbeta.data_ptr<float>(), dgamma.data_ptr<float>());
   DISPATCH_DT(x.scalar_type(), {
-    hipLaunchKernelGGL(bn_bwd_dx_kernel<dt>, dim3(blocks), dim3(kBlock), 0,
-                       stream,
+    hipLaunchKernelGGL(bn_bwd_dx_kernel<dt>, dim3(eblocks), dim3(kBlock),
+                       0, stream,
                        reinterpret_cast<const dt*>(dy.data_ptr()),
                        reinterpret_cast<const dt*>(x.data_ptr()),
                        reinterpret_cast<dt*>(dx.data_ptr()), g.rows, g.C,
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        gamma.defined() ? gamma.data_ptr<float>() : nullptr,
-                       partial.data_ptr<float>(), 1.f / (float)g.rows);
+                       dbeta.data_ptr<float>(), dgamma.data_ptr<float>(),
+                       1.f / (float)g.rows);
   });
   CHECK_HIP(hipGetLastError());
-  // dgamma = partial[C:2C], dbeta = partial[0:C]
-  auto dbeta = partial.narrow(0, 0, g.C).clone();
-  auto dgamma = partial.narrow(0, g.C, g.C).clone();
   return {dx, dgamma, dbeta};
 }
 
