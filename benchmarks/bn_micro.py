@@ -5,7 +5,8 @@ import sys
 import torch
 import torch.nn as nn
 
-sys.path.insert(0, '.')
+import os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from mgwfbp_amd.kernels.batchnorm import MgxBatchNorm2d  # noqa: E402
 
 # (C, H, W) with bs 128 — every distinct BN shape in resnet50
