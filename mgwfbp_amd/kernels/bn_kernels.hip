@@ -197,31 +197,59 @@ __global__ __launch_bounds__(kBlock) void bn_fwd_reduce_kernel(
                        partial);
 }
 
-// ------------------------------------------------ fwd: finalize ---------
-// Sum the per-block partials, then mean/invstd + running stats.
+// ------------------------------------------------ finalize --------------
+// Sum the [nblk][2C] partials. A first version used ceil(C/256) blocks
+// each looping all nblk rows: 1-8 blocks of latency-bound strided loads
+// = 200+ us, 3x the cost of the actual data passes. Now each block owns
+// kFinC channels and its 256 threads split the nblk rows 8 ways, with
+// an LDS combine — parallelism 8C instead of C.
+constexpr int kFinC = 32;                  // channels per finalize block
+constexpr int kFinG = kBlock / kFinC;      // row-subsets per block
+
+__device__ inline void finalize_sums(const float* __restrict__ partial,
+                                     long nblk, long C, long& c,
+                                     bool& leader, float& s, float& q) {
+  const int cl = threadIdx.x % kFinC;
+  const int g = threadIdx.x / kFinC;
+  c = (long)blockIdx.x * kFinC + cl;
+  s = 0.f; q = 0.f;
+  if (c < C) {
+    for (long b = g; b < nblk; b += kFinG) {
+      s += partial[b * 2 * C + c];
+      q += partial[b * 2 * C + C + c];
+    }
+  }
+  __shared__ float lds[kBlock * 2];
+  lds[threadIdx.x] = s;
+  lds[kBlock + threadIdx.x] = q;
+  __syncthreads();
+  leader = (g == 0 && c < C);
+  if (leader) {
+    for (int gg = 1; gg < kFinG; ++gg) {
+      s += lds[gg * kFinC + cl];
+      q += lds[kBlock + gg * kFinC + cl];
+    }
+  }
+}
+
 __global__ __launch_bounds__(kBlock) void bn_fwd_finalize_kernel(
     const float* __restrict__ partial, long nblk, long C, float M,
     float eps, float momentum, float* __restrict__ mean,
     float* __restrict__ invstd, float* __restrict__ running_mean,
     float* __restrict__ running_var) {
-  for (long c = blockIdx.x * blockDim.x + threadIdx.x; c < C;
-       c += (long)gridDim.x * blockDim.x) {
-    float s = 0.f, q = 0.f;
-    for (long b = 0; b < nblk; ++b) {
-      s += partial[b * 2 * C + c];
-      q += partial[b * 2 * C + C + c];
-    }
-    const float m = s / M;
-    float var = q / M - m * m;
-    var = var < 0.f ? 0.f : var;
-    mean[c] = m;
-    invstd[c] = rsqrtf(var + eps);
-    if (running_mean != nullptr) {
-      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
-      const float unbiased = M > 1.f ? var * M / (M - 1.f) : var;
-      running_var[c] =
-          (1.f - momentum) * running_var[c] + momentum * unbiased;
-    }
+  long c; bool leader; float s, q;
+  finalize_sums(partial, nblk, C, c, leader, s, q);
+  if (!leader) return;
+  const float m = s / M;
+  float var = q / M - m * m;
+  var = var < 0.f ? 0.f : var;
+  mean[c] = m;
+  invstd[c] = rsqrtf(var + eps);
+  if (running_mean != nullptr) {
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+    const float unbiased = M > 1.f ? var * M / (M - 1.f) : var;
+    running_var[c] =
+        (1.f - momentum) * running_var[c] + momentum * unbiased;
   }
 }
 
@@ -314,16 +342,11 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_reduce_kernel(
 __global__ __launch_bounds__(kBlock) void bn_bwd_finalize_kernel(
     const float* __restrict__ partial, long nblk, long C,
     float* __restrict__ dbeta, float* __restrict__ dgamma) {
-  for (long c = blockIdx.x * blockDim.x + threadIdx.x; c < C;
-       c += (long)gridDim.x * blockDim.x) {
-    float sd = 0.f, sx = 0.f;
-    for (long b = 0; b < nblk; ++b) {
-      sd += partial[b * 2 * C + c];
-      sx += partial[b * 2 * C + C + c];
-    }
-    dbeta[c] = sd;
-    dgamma[c] = sx;
-  }
+  long c; bool leader; float sd, sx;
+  finalize_sums(partial, nblk, C, c, leader, sd, sx);
+  if (!leader) return;
+  dbeta[c] = sd;
+  dgamma[c] = sx;
 }
 
 // ---------------------------------------------------- bwd: dx -----------
@@ -391,7 +414,8 @@ Geometry geom(const torch::Tensor& x) {
 // partials stays negligible.
 int grid_reduce(const Geometry& g) {
   long nblk = (g.rows + (long)g.rpb * 8 - 1) / ((long)g.rpb * 8);
-  if (nblk > 1024) nblk = 1024;
+  if (nblk > 320) nblk = 320;   // keeps the finalize pass tiny; 320*256
+                                // threads stream HBM fine at >=8 iters
   return (int)(nblk < 1 ? 1 : nblk);
 }
 
@@ -436,7 +460,7 @@ std::vector<torch::Tensor> bn_fwd_train(
                        g.C, partial.data_ptr<float>());
   });
   hipLaunchKernelGGL(bn_fwd_finalize_kernel,
-                     dim3((g.C + kBlock - 1) / kBlock), dim3(kBlock), 0,
+                     dim3((g.C + kFinC - 1) / kFinC), dim3(kBlock), 0,
                      stream, partial.data_ptr<float>(), rblocks, g.C,
                      (float)g.rows, (float)eps, (float)momentum,
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
@@ -506,7 +530,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                        partial.data_ptr<float>());
   });
   hipLaunchKernelGGL(bn_bwd_finalize_kernel,
-                     dim3((g.C + kBlock - 1) / kBlock), dim3(kBlock), 0,
+                     dim3((g.C + kFinC - 1) / kFinC), dim3(kBlock), 0,
                      stream, partial.data_ptr<float>(), rblocks, g.C,
                      dbeta.data_ptr<float>(), dgamma.data_ptr<float>());
   DISPATCH_DT(x.scalar_type(), {
