@@ -168,11 +168,28 @@ __global__ __launch_bounds__(kBlock) void bn_fwd_reduce_kernel(
   float q[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   long c0_last = 0;
   if (active) {
+    const long rstride = (long)gridDim.x * rpb;
     for (long ob = o; ob < octets; ob += opr) {
       const long c0 = ob * 8;
       c0_last = c0;
-      for (long r = (long)blockIdx.x * rpb + rl; r < rows;
-           r += (long)gridDim.x * rpb) {
+      long r = (long)blockIdx.x * rpb + rl;
+      // 4x unrolled: four independent 16B loads in flight per thread
+      // (the rolled loop compiles to load -> s_waitcnt vmcnt(0) -> use,
+      // one load outstanding = latency-bound)
+      for (; r + 3 * rstride < rows; r += 4 * rstride) {
+        float v0[8], v1[8], v2[8], v3[8];
+        VecIO<T>::load(x + r * C + c0, v0);
+        VecIO<T>::load(x + (r + rstride) * C + c0, v1);
+        VecIO<T>::load(x + (r + 2 * rstride) * C + c0, v2);
+        VecIO<T>::load(x + (r + 3 * rstride) * C + c0, v3);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          s[i] += v0[i] + v1[i] + v2[i] + v3[i];
+          q[i] += v0[i] * v0[i] + v1[i] * v1[i] + v2[i] * v2[i]
+                  + v3[i] * v3[i];
+        }
+      }
+      for (; r < rows; r += rstride) {
         float v[8];
         VecIO<T>::load(x + r * C + c0, v);
 #pragma unroll
@@ -273,8 +290,25 @@ __global__ __launch_bounds__(kBlock) void bn_fwd_norm_kernel(
       sc[i] = g * invstd[c0 + i];
       sh[i] = (beta ? beta[c0 + i] : 0.f) - mean[c0 + i] * sc[i];
     }
-    for (long r = (long)blockIdx.x * rpb + rl; r < rows;
-         r += (long)gridDim.x * rpb) {
+    const long rstride = (long)gridDim.x * rpb;
+    long r = (long)blockIdx.x * rpb + rl;
+    for (; r + rstride < rows; r += 2 * rstride) {
+      float v[8], w[8];
+      VecIO<T>::load(x + r * C + c0, v);
+      VecIO<T>::load(x + (r + rstride) * C + c0, w);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        v[i] = v[i] * sc[i] + sh[i];
+        w[i] = w[i] * sc[i] + sh[i];
+        if (relu) {
+          v[i] = v[i] > 0.f ? v[i] : 0.f;
+          w[i] = w[i] > 0.f ? w[i] : 0.f;
+        }
+      }
+      VecIO<T>::store(y + r * C + c0, v);
+      VecIO<T>::store(y + (r + rstride) * C + c0, w);
+    }
+    for (; r < rows; r += rstride) {
       float v[8];
       VecIO<T>::load(x + r * C + c0, v);
 #pragma unroll
@@ -310,8 +344,22 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_reduce_kernel(
         mu[i] = mean[c0 + i];
         is[i] = invstd[c0 + i];
       }
-      for (long r = (long)blockIdx.x * rpb + rl; r < rows;
-           r += (long)gridDim.x * rpb) {
+      const long rstride = (long)gridDim.x * rpb;
+      long r = (long)blockIdx.x * rpb + rl;
+      for (; r + rstride < rows; r += 2 * rstride) {  // 4 loads in flight
+        float g0[8], v0[8], g1[8], v1[8];
+        VecIO<T>::load(dy + r * C + c0, g0);
+        VecIO<T>::load(x + r * C + c0, v0);
+        VecIO<T>::load(dy + (r + rstride) * C + c0, g1);
+        VecIO<T>::load(x + (r + rstride) * C + c0, v1);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          sd[i] += g0[i] + g1[i];
+          sx[i] += g0[i] * (v0[i] - mu[i]) * is[i]
+                   + g1[i] * (v1[i] - mu[i]) * is[i];
+        }
+      }
+      for (; r < rows; r += rstride) {
         float g[8], v[8];
         VecIO<T>::load(dy + r * C + c0, g);
         VecIO<T>::load(x + r * C + c0, v);
@@ -373,8 +421,25 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_dx_kernel(
       md[i] = dbeta[c0 + i] * invM;
       mx[i] = dgamma[c0 + i] * invM;
     }
-    for (long r = (long)blockIdx.x * rpb + rl; r < rows;
-         r += (long)gridDim.x * rpb) {
+    const long rstride = (long)gridDim.x * rpb;
+    long r = (long)blockIdx.x * rpb + rl;
+    for (; r + rstride < rows; r += 2 * rstride) {
+      float g0[8], v0[8], g1[8], v1[8];
+      VecIO<T>::load(dy + r * C + c0, g0);
+      VecIO<T>::load(x + r * C + c0, v0);
+      VecIO<T>::load(dy + (r + rstride) * C + c0, g1);
+      VecIO<T>::load(x + (r + rstride) * C + c0, v1);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const float xh0 = (v0[i] - mu[i]) * is[i];
+        const float xh1 = (v1[i] - mu[i]) * is[i];
+        g0[i] = gi[i] * (g0[i] - md[i] - xh0 * mx[i]);
+        g1[i] = gi[i] * (g1[i] - md[i] - xh1 * mx[i]);
+      }
+      VecIO<T>::store(dx + r * C + c0, g0);
+      VecIO<T>::store(dx + (r + rstride) * C + c0, g1);
+    }
+    for (; r < rows; r += rstride) {
       float g[8], v[8];
       VecIO<T>::load(dy + r * C + c0, g);
       VecIO<T>::load(x + r * C + c0, v);
