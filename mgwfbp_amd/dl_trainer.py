@@ -155,7 +155,11 @@ class DLTrainer:
             and os.environ.get('MGX_CHANNELS_LAST', '1') == '1')
         if self.channels_last:
             self.net = self.net.to(memory_format=torch.channels_last)
-            if os.environ.get('MGX_FUSED_BN', '1') == '1':
+            # default off: measured at parity with MIOpen's NHWC BN on
+            # kernel time but ~25us/call extra host overhead (53 BN
+            # layers -> ~1.3 ms/step on resnet50); see
+            # profiles/resnet50_n1_bf16_kernel_stats.md + benchmarks/bn_micro.py
+            if os.environ.get('MGX_FUSED_BN', '0') == '1':
                 try:
                     from .kernels.batchnorm import convert_batchnorm
                     from .kernels import hip_available
