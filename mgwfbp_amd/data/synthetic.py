@@ -97,12 +97,22 @@ class SyntheticAudioDataset(Dataset):
         return spect, target
 
 
-def an4_collate(batch):
+def an4_collate(batch, pad_to=128):
     """Pad a batch of (spect, target) to the max T (reference
-    audio_data loader contract: padded batch + per-utterance lengths)."""
+    audio_data loader contract: padded batch + per-utterance lengths).
+
+    max_t is rounded up to a multiple of ``pad_to``: MIOpen's fused RNN
+    tunes per (T, N) shape, and un-quantized padding gives every batch a
+    fresh T — measured ~8 s of RNN re-tuning PER STEP on MI355X
+    (benchmarks/an4_probe.py). With a coarse grid only ~4 shapes exist
+    and each tunes once. CTC masks padded frames via output_sizes, so
+    numerics are unchanged.
+    """
     batch = sorted(batch, key=lambda b: b[0].size(2), reverse=True)
     freq = batch[0][0].size(1)
     max_t = batch[0][0].size(2)
+    if pad_to > 1:
+        max_t = (max_t + pad_to - 1) // pad_to * pad_to
     n = len(batch)
     inputs = torch.zeros(n, 1, freq, max_t)
     input_lengths = torch.zeros(n, dtype=torch.long)
