@@ -37,6 +37,11 @@ def main():
     parser.add_argument('--profile-iters', type=int, default=15,
                         help='layerwise-profiling iterations for the '
                              'solver (mgwfbp arm)')
+    parser.add_argument('--graph', type=str, default='auto',
+                        choices=['auto', 'on', 'off'],
+                        help='hipGraph-capture the training step (full '
+                             'compute incl. collectives; falls back to '
+                             'eager on capture failure)')
     args = parser.parse_args()
 
     world_size = int(os.environ.get('WORLD_SIZE', '1'))
@@ -107,16 +112,44 @@ def main():
     if comm.size() > 1:
         broadcast_parameters(trainer.net.state_dict(), root_rank=0)
 
-    def one_step():
+    def eager_step():
         optimizer.zero_grad()
         trainer.train(1)
         trainer.update_model()
 
+    one_step = eager_step
+    graphed = False
     phase('optimizer ready')
-    # warmup (untimed)
+    # warmup (untimed) — also tunes every MIOpen shape before capture
     for _ in range(args.warmup):
-        one_step()
+        eager_step()
     phase('warmup done (%d steps)' % args.warmup)
+
+    want_graph = (args.graph == 'on'
+                  or (args.graph == 'auto' and torch.cuda.is_available()))
+    if want_graph and torch.cuda.is_available():
+        from mgwfbp_amd.graph_step import GraphedTrainStep
+        try:
+            gstep = GraphedTrainStep(trainer, optimizer)
+            ok = torch.tensor([1.0], device=trainer.device)
+        except Exception as e:
+            print('[bench] graph capture failed (%s); eager path' % e,
+                  file=sys.stderr)
+            gstep = None
+            ok = torch.tensor([0.0], device=trainer.device)
+        # all ranks must agree before committing to the graph path
+        if comm.size() > 1:
+            h = comm.allreduce_async_(ok, average=False)
+            comm.synchronize(h)
+            torch.cuda.synchronize()
+        if gstep is not None and float(ok.item()) >= comm.size():
+            one_step = gstep.step
+            graphed = True
+            for _ in range(3):   # replay warmup
+                one_step()
+            phase('hipGraph captured (replay path)')
+        else:
+            phase('eager path (graph unavailable)')
 
     # timed region: barrier + sync on both sides
     comm.barrier()
@@ -168,6 +201,7 @@ def main():
                 'parallelism': 'dp%d' % n_gpus,
                 'merge': args.merge,
                 'comm_backend': comm.backend_name(),
+                'hip_graph': graphed,
             },
         }
         print(json.dumps(result))
