@@ -125,8 +125,12 @@ def main():
         eager_step()
     phase('warmup done (%d steps)' % args.warmup)
 
+    # auto: graph only at world==1 — RCCL collective capture at world>1
+    # is untested on this pool (single-GPU boxes); use --graph on to
+    # force it for multi-rank runs
     want_graph = (args.graph == 'on'
-                  or (args.graph == 'auto' and torch.cuda.is_available()))
+                  or (args.graph == 'auto' and torch.cuda.is_available()
+                      and n_gpus == 1))
     if want_graph and torch.cuda.is_available():
         from mgwfbp_amd.graph_step import GraphedTrainStep
         try:
