@@ -328,6 +328,13 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                 cbuf.copy_(flat.to(self._comm_dtype))
             handle = comm.allreduce_async_(cbuf, average=True, name=name)
         assert key not in self._handles
+        if settings.DETERMINISTIC:
+            # single-stream debug mode: complete each
+            # collective before backward continues — no overlap, no
+            # stream interleaving
+            handle.wait()
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
         self._handles[key] = handle
 
     # ------------------------------------------------------------------

@@ -16,12 +16,13 @@ def _data(seed, n=8):
     return x, y
 
 
-def _worker(rank, world, port, threshold, use_solver, q):
+def _worker(rank, world, port, threshold, use_solver, q, comm_dtype='fp32'):
     os.environ['MASTER_ADDR'] = '127.0.0.1'
     os.environ['MASTER_PORT'] = str(port)
     os.environ['WORLD_SIZE'] = str(world)
     os.environ['RANK'] = str(rank)
     os.environ['MGX_COMM_BACKEND'] = 'gloo'
+    os.environ['MGX_COMM_DTYPE'] = comm_dtype
     import mgwfbp_amd.comm as comm
     from mgwfbp_amd import models
     from mgwfbp_amd.distributed_optimizer import (DistributedOptimizer,
@@ -50,11 +51,12 @@ def _worker(rank, world, port, threshold, use_solver, q):
     comm.shutdown()
 
 
-def _run_world(world, threshold, use_solver, port):
+def _run_world(world, threshold, use_solver, port, comm_dtype='fp32'):
     ctx = mp.get_context('spawn')
     q = ctx.SimpleQueue()
     procs = [ctx.Process(target=_worker,
-                         args=(r, world, port, threshold, use_solver, q))
+                         args=(r, world, port, threshold, use_solver, q,
+                               comm_dtype))
              for r in range(world)]
     for p in procs:
         p.start()
@@ -99,4 +101,16 @@ def test_two_process_dp_matches_sequential(threshold, use_solver, port):
     # and match the sequential DP emulation
     ref = _reference_dp(world)
     assert torch.allclose(results[0], ref, atol=1e-5), \
+        (results[0] - ref).abs().max()
+
+
+def test_bf16_wire_format():
+    """COMM_DTYPE=bf16: gradients cross the wire as bf16 (half the
+    bytes); ranks must agree bitwise and track the fp32 DP reference to
+    bf16 precision."""
+    world = 2
+    results = _run_world(world, 0, False, 29617, comm_dtype='bf16')
+    assert torch.equal(results[0], results[1])
+    ref = _reference_dp(world)
+    assert torch.allclose(results[0], ref, atol=5e-2, rtol=5e-2), \
         (results[0] - ref).abs().max()
