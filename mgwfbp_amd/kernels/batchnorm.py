@@ -31,24 +31,28 @@ def _load():
 class _BatchNormFunc(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var, momentum,
-                eps):
+                eps, relu):
         ext = _load()
         y, mean, invstd = ext.bn_fwd_train(
             x, weight, bias, running_mean, running_var, momentum, eps,
-            False)
-        ctx.save_for_backward(x, weight, mean, invstd)
+            relu)
+        ctx.relu = relu
+        ctx.save_for_backward(x, weight, bias, mean, invstd)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         ext = _load()
-        x, weight, mean, invstd = ctx.saved_tensors
+        x, weight, bias, mean, invstd = ctx.saved_tensors
         dy = dy.contiguous(memory_format=torch.channels_last)
-        dx, dgamma, dbeta = ext.bn_bwd(dy, x, mean, invstd, weight)
-        return dx, dgamma, dbeta, None, None, None, None
+        dx, dgamma, dbeta = ext.bn_bwd(dy, x, mean, invstd, weight, bias,
+                                       ctx.relu)
+        return dx, dgamma, dbeta, None, None, None, None, None
 
 
 class MgxBatchNorm2d(nn.BatchNorm2d):
+    fuse_relu = False
+
     def _use_hip_path(self, x):
         return (x.is_cuda and x.dim() == 4 and x.size(1) % 8 == 0
                 and x.is_contiguous(memory_format=torch.channels_last)
@@ -57,7 +61,10 @@ class MgxBatchNorm2d(nn.BatchNorm2d):
 
     def forward(self, x):
         if not self._use_hip_path(x):
-            return super().forward(x)
+            y = super().forward(x)
+            if self.fuse_relu:
+                y = torch.nn.functional.relu(y, inplace=True)
+            return y
         if self.training:
             if self.num_batches_tracked is not None:
                 self.num_batches_tracked.add_(1)
@@ -67,29 +74,48 @@ class MgxBatchNorm2d(nn.BatchNorm2d):
             return _BatchNormFunc.apply(x, self.weight, self.bias,
                                         self.running_mean,
                                         self.running_var, momentum,
-                                        self.eps)
+                                        self.eps, self.fuse_relu)
         ext = _load()
         return ext.bn_fwd_eval(x, self.weight, self.bias,
                                self.running_mean, self.running_var,
-                               self.eps, False)
+                               self.eps, self.fuse_relu)
 
 
-def convert_batchnorm(module):
+def _mgx_from(child, fuse_relu=False):
+    bn = MgxBatchNorm2d(child.num_features, eps=child.eps,
+                        momentum=child.momentum, affine=child.affine,
+                        track_running_stats=child.track_running_stats)
+    bn = bn.to(device=child.weight.device if child.affine else 'cpu',
+               dtype=child.weight.dtype if child.affine
+               else torch.float32)
+    bn.load_state_dict(child.state_dict())
+    bn.fuse_relu = fuse_relu
+    return bn
+
+
+def convert_batchnorm(module, fuse_relu=True):
     """Recursively swap nn.BatchNorm2d -> MgxBatchNorm2d (in place),
-    keeping parameters, buffers and config."""
-    for name, child in module.named_children():
-        if type(child) is nn.BatchNorm2d:
-            bn = MgxBatchNorm2d(child.num_features, eps=child.eps,
-                                momentum=child.momentum,
-                                affine=child.affine,
-                                track_running_stats=
-                                child.track_running_stats)
-            bn = bn.to(device=child.weight.device
-                       if child.affine else 'cpu',
-                       dtype=child.weight.dtype if child.affine
-                       else torch.float32)
-            bn.load_state_dict(child.state_dict())
-            setattr(module, name, bn)
+    keeping parameters, buffers and config.
+
+    With ``fuse_relu``, BN+ReLU pairs become ONE fused op:
+    - a ``models.common.BNReLU`` unit gets a relu-fused MgxBatchNorm2d;
+    - inside an nn.Sequential, a (BatchNorm2d, ReLU) pair becomes
+      (fused MgxBatchNorm2d, Identity) — covers VGG-style stacks.
+    """
+    from ..models.common import BNReLU
+    children = list(module.named_children())
+    names = [n for n, _ in children]
+    for idx, (name, child) in enumerate(children):
+        if isinstance(child, BNReLU):
+            if type(child.bn) is nn.BatchNorm2d:
+                child.bn = _mgx_from(child.bn, fuse_relu=fuse_relu)
+        elif type(child) is nn.BatchNorm2d:
+            relu_next = (fuse_relu and isinstance(module, nn.Sequential)
+                         and idx + 1 < len(children)
+                         and isinstance(children[idx + 1][1], nn.ReLU))
+            setattr(module, name, _mgx_from(child, fuse_relu=relu_next))
+            if relu_next:
+                setattr(module, names[idx + 1], nn.Identity())
         else:
-            convert_batchnorm(child)
+            convert_batchnorm(child, fuse_relu=fuse_relu)
     return module

@@ -322,10 +322,14 @@ __global__ __launch_bounds__(kBlock) void bn_fwd_norm_kernel(
 }
 
 // -------------------------------------------------- bwd: reduce ---------
-template <typename T>
+// RELU: the forward fused y = relu(bn(x)); backward gates dy by y>0,
+// recomputing y's sign from (x, mean, invstd, gamma, beta) — no saved
+// activation needed.
+template <typename T, bool RELU>
 __global__ __launch_bounds__(kBlock) void bn_bwd_reduce_kernel(
     const T* __restrict__ dy, const T* __restrict__ x, long rows, long C,
     const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
     float* __restrict__ partial) {
   int opr, rpb, o, rl; bool active;
   tile_map(C, opr, rpb, o, rl, active);
@@ -338,11 +342,15 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_reduce_kernel(
     for (long ob = o; ob < octets; ob += opr) {
       const long c0 = ob * 8;
       c0_last = c0;
-      float mu[8], is[8];
+      float mu[8], is[8], ga[8], be[8];
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         mu[i] = mean[c0 + i];
         is[i] = invstd[c0 + i];
+        if (RELU) {
+          ga[i] = gamma ? gamma[c0 + i] : 1.f;
+          be[i] = beta ? beta[c0 + i] : 0.f;
+        }
       }
       const long rstride = (long)gridDim.x * rpb;
       long r = (long)blockIdx.x * rpb + rl;
@@ -354,9 +362,14 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_reduce_kernel(
         VecIO<T>::load(x + (r + rstride) * C + c0, v1);
 #pragma unroll
         for (int i = 0; i < 8; ++i) {
+          const float xh0 = (v0[i] - mu[i]) * is[i];
+          const float xh1 = (v1[i] - mu[i]) * is[i];
+          if (RELU) {
+            if (xh0 * ga[i] + be[i] <= 0.f) g0[i] = 0.f;
+            if (xh1 * ga[i] + be[i] <= 0.f) g1[i] = 0.f;
+          }
           sd[i] += g0[i] + g1[i];
-          sx[i] += g0[i] * (v0[i] - mu[i]) * is[i]
-                   + g1[i] * (v1[i] - mu[i]) * is[i];
+          sx[i] += g0[i] * xh0 + g1[i] * xh1;
         }
       }
       for (; r < rows; r += rstride) {
@@ -366,6 +379,7 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_reduce_kernel(
 #pragma unroll
         for (int i = 0; i < 8; ++i) {
           const float xh = (v[i] - mu[i]) * is[i];
+          if (RELU && xh * ga[i] + be[i] <= 0.f) g[i] = 0.f;
           sd[i] += g[i];
           sx[i] += g[i] * xh;
         }
@@ -399,25 +413,27 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_finalize_kernel(
 
 // ---------------------------------------------------- bwd: dx -----------
 // dx = gamma*invstd*(dy - dbeta/M - xhat*dgamma/M)
-template <typename T>
+template <typename T, bool RELU>
 __global__ __launch_bounds__(kBlock) void bn_bwd_dx_kernel(
     const T* __restrict__ dy, const T* __restrict__ x, T* __restrict__ dx,
     long rows, long C, const float* __restrict__ mean,
     const float* __restrict__ invstd, const float* __restrict__ gamma,
-    const float* __restrict__ dbeta, const float* __restrict__ dgamma,
-    float invM) {
+    const float* __restrict__ beta, const float* __restrict__ dbeta,
+    const float* __restrict__ dgamma, float invM) {
   int opr, rpb, o, rl; bool active;
   tile_map(C, opr, rpb, o, rl, active);
   if (!active) return;
   const long octets = C / 8;
   for (long ob = o; ob < octets; ob += opr) {
     const long c0 = ob * 8;
-    float mu[8], is[8], gi[8], md[8], mx[8];
+    float mu[8], is[8], gi[8], md[8], mx[8], ga[8], be[8];
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
       mu[i] = mean[c0 + i];
       is[i] = invstd[c0 + i];
-      gi[i] = (gamma ? gamma[c0 + i] : 1.f) * is[i];
+      ga[i] = gamma ? gamma[c0 + i] : 1.f;
+      be[i] = beta ? beta[c0 + i] : 0.f;
+      gi[i] = ga[i] * is[i];
       md[i] = dbeta[c0 + i] * invM;
       mx[i] = dgamma[c0 + i] * invM;
     }
@@ -433,6 +449,10 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_dx_kernel(
       for (int i = 0; i < 8; ++i) {
         const float xh0 = (v0[i] - mu[i]) * is[i];
         const float xh1 = (v1[i] - mu[i]) * is[i];
+        if (RELU) {
+          if (xh0 * ga[i] + be[i] <= 0.f) g0[i] = 0.f;
+          if (xh1 * ga[i] + be[i] <= 0.f) g1[i] = 0.f;
+        }
         g0[i] = gi[i] * (g0[i] - md[i] - xh0 * mx[i]);
         g1[i] = gi[i] * (g1[i] - md[i] - xh1 * mx[i]);
       }
@@ -446,6 +466,7 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_dx_kernel(
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         const float xh = (v[i] - mu[i]) * is[i];
+        if (RELU && xh * ga[i] + be[i] <= 0.f) g[i] = 0.f;
         g[i] = gi[i] * (g[i] - md[i] - xh * mx[i]);
       }
       VecIO<T>::store(dx + r * C + c0, g);
@@ -573,7 +594,8 @@ torch::Tensor bn_fwd_eval(torch::Tensor x, torch::Tensor gamma,
 
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor mean, torch::Tensor invstd,
-                                  torch::Tensor gamma) {
+                                  torch::Tensor gamma, torch::Tensor beta,
+                                  bool relu) {
   auto g = geom(x);
   TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast),
               "expected channels_last grad");
@@ -585,29 +607,50 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   auto dbeta = torch::empty({g.C}, opts);
   auto dgamma = torch::empty({g.C}, opts);
   auto dx = torch::empty_like(x);
+  const float* gamma_p = gamma.defined() ? gamma.data_ptr<float>()
+                                         : nullptr;
+  const float* beta_p = beta.defined() ? beta.data_ptr<float>() : nullptr;
   DISPATCH_DT(x.scalar_type(), {
-    hipLaunchKernelGGL(bn_bwd_reduce_kernel<dt>, dim3(rblocks),
-                       dim3(kBlock), 0, stream,
-                       reinterpret_cast<const dt*>(dy.data_ptr()),
-                       reinterpret_cast<const dt*>(x.data_ptr()), g.rows,
-                       g.C, mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(),
-                       partial.data_ptr<float>());
+    if (relu)
+      hipLaunchKernelGGL((bn_bwd_reduce_kernel<dt, true>), dim3(rblocks),
+                         dim3(kBlock), 0, stream,
+                         reinterpret_cast<const dt*>(dy.data_ptr()),
+                         reinterpret_cast<const dt*>(x.data_ptr()), g.rows,
+                         g.C, mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(), gamma_p, beta_p,
+                         partial.data_ptr<float>());
+    else
+      hipLaunchKernelGGL((bn_bwd_reduce_kernel<dt, false>), dim3(rblocks),
+                         dim3(kBlock), 0, stream,
+                         reinterpret_cast<const dt*>(dy.data_ptr()),
+                         reinterpret_cast<const dt*>(x.data_ptr()), g.rows,
+                         g.C, mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(), gamma_p, beta_p,
+                         partial.data_ptr<float>());
   });
   hipLaunchKernelGGL(bn_bwd_finalize_kernel,
                      dim3((g.C + kFinC - 1) / kFinC), dim3(kBlock), 0,
                      stream, partial.data_ptr<float>(), rblocks, g.C,
                      dbeta.data_ptr<float>(), dgamma.data_ptr<float>());
   DISPATCH_DT(x.scalar_type(), {
-    hipLaunchKernelGGL(bn_bwd_dx_kernel<dt>, dim3(eblocks), dim3(kBlock),
-                       0, stream,
-                       reinterpret_cast<const dt*>(dy.data_ptr()),
-                       reinterpret_cast<const dt*>(x.data_ptr()),
-                       reinterpret_cast<dt*>(dx.data_ptr()), g.rows, g.C,
-                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       gamma.defined() ? gamma.data_ptr<float>() : nullptr,
-                       dbeta.data_ptr<float>(), dgamma.data_ptr<float>(),
-                       1.f / (float)g.rows);
+    if (relu)
+      hipLaunchKernelGGL((bn_bwd_dx_kernel<dt, true>), dim3(eblocks),
+                         dim3(kBlock), 0, stream,
+                         reinterpret_cast<const dt*>(dy.data_ptr()),
+                         reinterpret_cast<const dt*>(x.data_ptr()),
+                         reinterpret_cast<dt*>(dx.data_ptr()), g.rows, g.C,
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         gamma_p, beta_p, dbeta.data_ptr<float>(),
+                         dgamma.data_ptr<float>(), 1.f / (float)g.rows);
+    else
+      hipLaunchKernelGGL((bn_bwd_dx_kernel<dt, false>), dim3(eblocks),
+                         dim3(kBlock), 0, stream,
+                         reinterpret_cast<const dt*>(dy.data_ptr()),
+                         reinterpret_cast<const dt*>(x.data_ptr()),
+                         reinterpret_cast<dt*>(dx.data_ptr()), g.rows, g.C,
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         gamma_p, beta_p, dbeta.data_ptr<float>(),
+                         dgamma.data_ptr<float>(), 1.f / (float)g.rows);
   });
   CHECK_HIP(hipGetLastError());
   return {dx, dgamma, dbeta};

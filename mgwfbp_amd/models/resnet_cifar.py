@@ -10,6 +10,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from .common import BNReLU
+
 
 def _conv3x3(cin, cout, stride=1):
     return nn.Conv2d(cin, cout, kernel_size=3, stride=stride, padding=1,
@@ -35,9 +37,9 @@ class BasicBlock(nn.Module):
     def __init__(self, cin, cout, stride=1, option='B'):
         super().__init__()
         self.conv1 = _conv3x3(cin, cout, stride)
-        self.bn1 = nn.BatchNorm2d(cout)
+        self.bn1 = BNReLU(cout)
         self.conv2 = _conv3x3(cout, cout)
-        self.bn2 = nn.BatchNorm2d(cout)
+        self.bn2 = nn.BatchNorm2d(cout)   # relu after the add
         self.shortcut = nn.Sequential()
         if stride != 1 or cin != cout:
             if option == 'A':
@@ -49,7 +51,7 @@ class BasicBlock(nn.Module):
                     nn.BatchNorm2d(cout))
 
     def forward(self, x):
-        out = F.relu(self.bn1(self.conv1(x)))
+        out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
         out = out + self.shortcut(x)
         return F.relu(out)
@@ -61,7 +63,7 @@ class CifarResNet(nn.Module):
         assert (depth - 2) % 6 == 0, 'depth must be 6n+2'
         n = (depth - 2) // 6
         self.conv1 = _conv3x3(3, 16)
-        self.bn1 = nn.BatchNorm2d(16)
+        self.bn1 = BNReLU(16)
         self.layer1 = self._make_layer(16, 16, n, 1, option)
         self.layer2 = self._make_layer(16, 32, n, 2, option)
         self.layer3 = self._make_layer(32, 64, n, 2, option)
@@ -81,7 +83,7 @@ class CifarResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        out = F.relu(self.bn1(self.conv1(x)))
+        out = self.bn1(self.conv1(x))
         out = self.layer3(self.layer2(self.layer1(out)))
         out = F.adaptive_avg_pool2d(out, 1).flatten(1)
         return self.fc(out)

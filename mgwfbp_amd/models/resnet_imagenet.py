@@ -7,6 +7,8 @@ torchvision is not part of this framework's dependency set.
 import torch.nn as nn
 import torch.nn.functional as F
 
+from .common import BNReLU
+
 
 class BasicBlock(nn.Module):
     expansion = 1
@@ -14,14 +16,14 @@ class BasicBlock(nn.Module):
     def __init__(self, cin, width, stride=1, downsample=None):
         super().__init__()
         self.conv1 = nn.Conv2d(cin, width, 3, stride, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(width)
+        self.bn1 = BNReLU(width)
         self.conv2 = nn.Conv2d(width, width, 3, 1, 1, bias=False)
-        self.bn2 = nn.BatchNorm2d(width)
+        self.bn2 = nn.BatchNorm2d(width)   # relu comes after the add
         self.downsample = downsample
 
     def forward(self, x):
         identity = x if self.downsample is None else self.downsample(x)
-        out = F.relu(self.bn1(self.conv1(x)), inplace=True)
+        out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
         return F.relu(out + identity, inplace=True)
 
@@ -32,17 +34,17 @@ class Bottleneck(nn.Module):
     def __init__(self, cin, width, stride=1, downsample=None):
         super().__init__()
         self.conv1 = nn.Conv2d(cin, width, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(width)
+        self.bn1 = BNReLU(width)
         self.conv2 = nn.Conv2d(width, width, 3, stride, 1, bias=False)
-        self.bn2 = nn.BatchNorm2d(width)
+        self.bn2 = BNReLU(width)
         self.conv3 = nn.Conv2d(width, width * 4, 1, bias=False)
-        self.bn3 = nn.BatchNorm2d(width * 4)
+        self.bn3 = nn.BatchNorm2d(width * 4)   # relu after the add
         self.downsample = downsample
 
     def forward(self, x):
         identity = x if self.downsample is None else self.downsample(x)
-        out = F.relu(self.bn1(self.conv1(x)), inplace=True)
-        out = F.relu(self.bn2(self.conv2(out)), inplace=True)
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
         out = self.bn3(self.conv3(out))
         return F.relu(out + identity, inplace=True)
 
@@ -52,7 +54,7 @@ class ResNet(nn.Module):
         super().__init__()
         self.inplanes = 64
         self.conv1 = nn.Conv2d(3, 64, 7, 2, 3, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
+        self.bn1 = BNReLU(64)
         self.maxpool = nn.MaxPool2d(3, 2, 1)
         self.layer1 = self._make_layer(block, 64, layers[0])
         self.layer2 = self._make_layer(block, 128, layers[1], 2)
@@ -81,7 +83,7 @@ class ResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        x = F.relu(self.bn1(self.conv1(x)), inplace=True)
+        x = self.bn1(self.conv1(x))
         x = self.maxpool(x)
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         x = F.adaptive_avg_pool2d(x, 1).flatten(1)

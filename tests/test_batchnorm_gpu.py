@@ -86,3 +86,59 @@ def test_convert_resnet50_trains(capsys):
     loss.backward()
     torch.cuda.synchronize()
     assert torch.isfinite(loss)
+
+
+@pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16])
+def test_fused_bn_relu_matches_composite(dtype):
+    """y = relu(bn(x)) fused: forward + gated backward vs the fp32
+    composite torch reference."""
+    from mgwfbp_amd.kernels.batchnorm import MgxBatchNorm2d
+    C = 128
+    torch.manual_seed(3)
+    ref = nn.BatchNorm2d(C).cuda()
+    ref.weight.data.uniform_(0.5, 1.5)
+    ref.bias.data.uniform_(-0.5, 0.5)
+    ours = MgxBatchNorm2d(C).cuda()
+    ours.load_state_dict(ref.state_dict())
+    ours.fuse_relu = True
+
+    x1 = _mk(C, dtype=dtype, seed=7).requires_grad_(True)
+    x2 = x1.detach().float().clone().requires_grad_(True)
+    y_ref = torch.relu(ref(x2))
+    y = ours(x1)
+    tol = 2e-5 if dtype == torch.float32 else 5e-2
+    assert torch.allclose(y.float(), y_ref, atol=tol, rtol=tol)
+    assert (y.float().min() >= 0)
+
+    dy = torch.randn_like(y_ref)
+    y_ref.backward(dy)
+    y.backward(dy.to(dtype).to(memory_format=torch.channels_last))
+    assert torch.allclose(x1.grad.float(), x2.grad, atol=tol * 5,
+                          rtol=tol * 5), \
+        (x1.grad.float() - x2.grad).abs().max().item()
+    gatol, grtol = (1e-2, 1e-3) if dtype == torch.float32 else (0.5, 3e-2)
+    assert torch.allclose(ours.weight.grad, ref.weight.grad, atol=gatol,
+                          rtol=grtol)
+    assert torch.allclose(ours.bias.grad, ref.bias.grad, atol=gatol,
+                          rtol=grtol)
+
+
+def test_convert_fuses_bnrelu_units_and_sequentials():
+    from mgwfbp_amd import models
+    from mgwfbp_amd.kernels.batchnorm import (convert_batchnorm,
+                                              MgxBatchNorm2d)
+    net = models.resnet50().cuda().to(memory_format=torch.channels_last)
+    convert_batchnorm(net)
+    fused = sum(1 for m in net.modules()
+                if isinstance(m, MgxBatchNorm2d) and m.fuse_relu)
+    plain = sum(1 for m in net.modules()
+                if isinstance(m, MgxBatchNorm2d) and not m.fuse_relu)
+    # stem + 2 per bottleneck x16 = 33 fused; bn3 x16 + 4 downsample = 20
+    assert fused == 33, fused
+    assert plain == 20, plain
+    # vgg-style sequential fusion
+    vgg = models.vgg16().cuda().to(memory_format=torch.channels_last)
+    convert_batchnorm(vgg)
+    fused_v = sum(1 for m in vgg.modules()
+                  if isinstance(m, MgxBatchNorm2d) and m.fuse_relu)
+    assert fused_v == 13, fused_v
