@@ -155,16 +155,18 @@ class DLTrainer:
             and os.environ.get('MGX_CHANNELS_LAST', '1') == '1')
         if self.channels_last:
             self.net = self.net.to(memory_format=torch.channels_last)
-            # default off: measured at parity with MIOpen's NHWC BN on
-            # kernel time but ~25us/call extra host overhead (53 BN
-            # layers -> ~1.3 ms/step on resnet50); see
-            # profiles/resnet50_n1_bf16_kernel_stats.md + benchmarks/bn_micro.py
-            if os.environ.get('MGX_FUSED_BN', '0') == '1':
+            # MGX_FUSED_BN: '0' = MIOpen BN everywhere; 'relu' = fuse
+            # only BN+ReLU pairs into the MgxBatchNorm2d kernel, plain
+            # BNs stay on MIOpen (measured best mix on big models);
+            # '1' = convert every BN. See profiles/README.md for the A/B.
+            mode = os.environ.get('MGX_FUSED_BN', '0')
+            if mode in ('1', 'relu'):
                 try:
                     from .kernels.batchnorm import convert_batchnorm
                     from .kernels import hip_available
                     if hip_available():
-                        convert_batchnorm(self.net)
+                        convert_batchnorm(self.net,
+                                          only_fused=(mode == 'relu'))
                 except Exception as e:
                     logger.warning('fused BN unavailable: %s', e)
         if settings.DEBUG and rank == 0:

@@ -93,7 +93,7 @@ def _mgx_from(child, fuse_relu=False):
     return bn
 
 
-def convert_batchnorm(module, fuse_relu=True):
+def convert_batchnorm(module, fuse_relu=True, only_fused=False):
     """Recursively swap nn.BatchNorm2d -> MgxBatchNorm2d (in place),
     keeping parameters, buffers and config.
 
@@ -101,21 +101,30 @@ def convert_batchnorm(module, fuse_relu=True):
     - a ``models.common.BNReLU`` unit gets a relu-fused MgxBatchNorm2d;
     - inside an nn.Sequential, a (BatchNorm2d, ReLU) pair becomes
       (fused MgxBatchNorm2d, Identity) — covers VGG-style stacks.
+
+    ``only_fused`` leaves plain (no-relu) BatchNorm2d on the MIOpen path
+    and converts only the fusable pairs — MIOpen's NHWC BN is slightly
+    faster per-kernel on large layers, so converting everything costs
+    more than the relu fusion saves on big models (measured A/B,
+    profiles/README.md).
     """
     from ..models.common import BNReLU
     children = list(module.named_children())
     names = [n for n, _ in children]
     for idx, (name, child) in enumerate(children):
         if isinstance(child, BNReLU):
-            if type(child.bn) is nn.BatchNorm2d:
-                child.bn = _mgx_from(child.bn, fuse_relu=fuse_relu)
+            if type(child.bn) is nn.BatchNorm2d and fuse_relu:
+                child.bn = _mgx_from(child.bn, fuse_relu=True)
         elif type(child) is nn.BatchNorm2d:
             relu_next = (fuse_relu and isinstance(module, nn.Sequential)
                          and idx + 1 < len(children)
                          and isinstance(children[idx + 1][1], nn.ReLU))
-            setattr(module, name, _mgx_from(child, fuse_relu=relu_next))
             if relu_next:
+                setattr(module, name, _mgx_from(child, fuse_relu=True))
                 setattr(module, names[idx + 1], nn.Identity())
+            elif not only_fused:
+                setattr(module, name, _mgx_from(child, fuse_relu=False))
         else:
-            convert_batchnorm(child, fuse_relu=fuse_relu)
+            convert_batchnorm(child, fuse_relu=fuse_relu,
+                              only_fused=only_fused)
     return module
