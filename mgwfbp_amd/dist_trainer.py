@@ -39,7 +39,7 @@ def mgwfbp(dnn, dataset, data_dir, nworkers, lr, batch_size, nsteps_update,
                         is_weak_scaling=True, ngpus=1, data_dir=data_dir,
                         dataset=dataset, dnn=dnn, lr=lr, nworkers=nworkers,
                         prefix=settings.PREFIX, pretrain=pretrain,
-                        num_steps=num_steps)
+                        num_steps=num_steps, tb_writer=writer)
     init_epoch = torch.tensor([trainer.train_epoch], dtype=torch.int64)
     init_iter = torch.tensor([trainer.train_iter], dtype=torch.int64)
     if comm.size() > 1:
@@ -148,12 +148,21 @@ def main():
     settings.add_file_handler(os.path.join(
         logdir, '%s-rank%d.log' % (settings.hostname, comm.rank())))
     logger.info('configurations: %s', args)
+    writer = None
+    if settings.TENSORBOARD and comm.rank() == 0:
+        try:
+            from torch.utils.tensorboard import SummaryWriter
+            writer = SummaryWriter(log_dir=logdir)
+        except ImportError:
+            logger.warning('MGX_TENSORBOARD set but the tensorboard '
+                           'package is not installed')
     mgwfbp(args.dnn, args.dataset, args.data_dir, nworkers, args.lr,
            args.batch_size, args.nsteps_update, args.max_epochs,
            nwpernode=args.nwpernode, pretrain=args.pretrain,
            num_steps=args.num_steps, compressor=args.compressor,
            density=args.density, threshold=args.threshold,
-           save_epochs=args.save_epochs, max_iters=args.max_iters)
+           writer=writer, save_epochs=args.save_epochs,
+           max_iters=args.max_iters)
     comm.shutdown()
 
 

@@ -360,6 +360,24 @@ class _DistributedOptimizer(torch.optim.Optimizer):
             self._groups_flags[gi] = [0] * len(self._groups_flags[gi])
         self._handles.clear()
         self.train_iter += 1
+        self._print_profiling()
+
+    def _print_profiling(self):
+        """Every 40 profiled iterations, log mean per-group allreduce
+        wait time (reference distributed_optimizer.py:407-425)."""
+        if not (self._profiling and comm.rank() == 0
+                and self._allreduce_timers):
+            return
+        first = next(iter(self._allreduce_timers.values()))
+        if len(first) != 40:
+            return
+        total = 0.0
+        for k, samples in self._allreduce_timers.items():
+            total += sum(samples) / len(samples)
+        logger.info('[%d]: mean allreduce wait per iter: %f s over %d '
+                    'groups', comm.rank(), total,
+                    len(self._allreduce_timers))
+        self._allreduce_timers.clear()
 
     def _clip_merged(self, flat):
         # reference :380-389: per-merged-tensor L2 clip at

@@ -470,6 +470,10 @@ class DLTrainer:
             self.train_iter += 1
         self.times.append(time.time() - s)
 
+        if self.writer is not None and self.rank == 0:
+            self.writer.add_scalar('train/loss', loss_value,
+                                   self.train_iter)
+            self.writer.add_scalar('train/lr', self.lr, self.train_iter)
         if self.train_iter % self.display == 0 and self.rank == 0:
             n = min(len(self.times), self.display)
             avg = sum(self.times[-n:]) / n
