@@ -20,10 +20,8 @@ from .settings import logger
 from .compression import compressors
 from .dl_trainer import DLTrainer
 from .distributed_optimizer import (DistributedOptimizer,
-                                    broadcast_parameters,
-                                    broadcast_optimizer_state)
+                                    broadcast_parameters)
 from .profiling import benchmark
-from . import models
 
 
 def mgwfbp(dnn, dataset, data_dir, nworkers, lr, batch_size, nsteps_update,

@@ -16,11 +16,8 @@ same reference implementations back the kernel numerics tests.
 from __future__ import annotations
 
 import importlib
-import os
 
 import torch
-
-from ..settings import logger
 
 _kernels_ext = None
 _comm_ext = None

@@ -2,8 +2,6 @@
 cost model lives in solver.py here)."""
 import os
 
-import torch
-
 
 def create_path(path):
     os.makedirs(path, exist_ok=True)
