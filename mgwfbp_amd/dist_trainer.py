@@ -87,19 +87,48 @@ def mgwfbp(dnn, dataset, data_dir, nworkers, lr, batch_size, nsteps_update,
     hidden = trainer.net.init_hidden() if dnn == 'lstm' else None
     nupdates = (max_iters if max_iters is not None
                 else max_epochs * iters_per_epoch)
-    for i in range(trainer.train_iter, nupdates):
+    # MGX_HIP_GRAPH=1: capture the step as a hipGraph after a few eager
+    # iterations and replay it (LR schedule flows through the fused
+    # SGD's device LR buffer). RNN workloads and gradient accumulation
+    # stay eager.
+    want_graph = (os.environ.get('MGX_HIP_GRAPH', '0') == '1'
+                  and torch.cuda.is_available()
+                  and dnn not in ('lstm', 'lstman4')
+                  and nsteps_update == 1)
+    gstep = None
+    start_iter = trainer.train_iter
+    for i in range(start_iter, nupdates):
         s = time.time()
-        optimizer.zero_grad()
-        for j in range(nsteps_update):
-            optimizer.local = (j < nsteps_update - 1)
-            _, hidden = trainer.train(1, hidden=hidden)
-        if dnn in ('lstm', 'lstman4'):
-            # external clipping path (reference dist_trainer.py:89-94)
-            optimizer.synchronize()
-            torch.nn.utils.clip_grad_norm_(trainer.net.parameters(),
-                                           norm_clip)
-        trainer.update_model()
-        times.append(time.time() - s)
+        if want_graph and gstep is None and i >= start_iter + 3:
+            try:
+                from .graph_step import GraphedTrainStep
+                gstep = GraphedTrainStep(trainer, optimizer)
+                trainer.train_iter += 2   # capture warmup ran 2 steps
+            except Exception as e:
+                logger.warning('hipGraph capture failed (%s); staying '
+                               'eager', e)
+                want_graph = False
+        if gstep is not None:
+            lr = trainer.adjust_learning_rate(trainer.train_epoch,
+                                              optimizer)
+            gstep.set_lr(lr)
+            gstep.step()
+            trainer.train_iter += 1
+            if trainer.train_iter % iters_per_epoch == 0:
+                trainer.train_epoch += 1
+            times.append(time.time() - s)
+        else:
+            optimizer.zero_grad()
+            for j in range(nsteps_update):
+                optimizer.local = (j < nsteps_update - 1)
+                _, hidden = trainer.train(1, hidden=hidden)
+            if dnn in ('lstm', 'lstman4'):
+                # external clipping path (reference dist_trainer.py:89-94)
+                optimizer.synchronize()
+                torch.nn.utils.clip_grad_norm_(trainer.net.parameters(),
+                                               norm_clip)
+            trainer.update_model()
+            times.append(time.time() - s)
         if i % display == 0 and i > 0 and rank == 0:
             avg = sum(times[-display:]) / min(len(times), display)
             logger.info('Time per iteration including communication: %f, '

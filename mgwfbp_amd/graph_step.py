@@ -13,9 +13,10 @@ This is the MI355X answer to a tracing compiler: HIP streams provide
 the comm/compute overlap, hipGraphs remove the launch overhead
 (BASELINE.json north star).
 
-Caveats: the learning rate is baked into the captured SGD launch
-(re-capture after LR changes — the benchmark holds LR constant); loss
-values are read from the static loss tensor after replay when needed.
+The fused SGD reads its learning rate from a 1-element device buffer,
+so the LR schedule keeps working across replays (``set_lr`` before each
+``step``) — no re-capture needed. Loss values are read from the static
+loss tensor after replay when needed.
 """
 from __future__ import annotations
 
@@ -37,6 +38,12 @@ class GraphedTrainStep:
         self.static_x = x.clone()
         self.static_y = y.clone()
         self.static_loss = None
+        # LR through a device buffer: replays follow the LR schedule
+        # without re-capture (set_lr fills it before replay)
+        self._fused = getattr(optimizer, '_fused_sgd', None)
+        if self._fused is not None:
+            self._fused.enable_lr_buffer()
+            self._fused.set_lr(optimizer.param_groups[0]['lr'])
 
         # side-stream warmup, then capture (torch.cuda.graph idiom)
         side = torch.cuda.Stream()
@@ -61,6 +68,10 @@ class GraphedTrainStep:
         loss.backward()
         self.trainer.update_model()
         self.static_loss = loss.detach()
+
+    def set_lr(self, lr):
+        if self._fused is not None:
+            self._fused.set_lr(lr)
 
     def step(self):
         x, y = self.trainer.fetch_data()

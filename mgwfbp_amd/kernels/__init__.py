@@ -102,12 +102,29 @@ class FusedSGD:
         self._t_tensors = t_tensors
         self._t_chunks = t_chunks
         self._nchunks = int(t_n.item())
+        self._lr_buf = torch.empty(0)   # disabled by default
+
+    def enable_lr_buffer(self):
+        """Switch the kernel to read LR from a device buffer — required
+        under hipGraph capture so the LR schedule survives replay (the
+        host fills the buffer before each replay, outside the graph)."""
+        if self._lr_buf.numel() == 0:
+            dev = self._t_tensors.device
+            self._lr_buf = torch.zeros(1, dtype=torch.float32, device=dev)
+        return self._lr_buf
+
+    def set_lr(self, lr):
+        self._lr_buf.fill_(float(lr))
 
     def step(self, lr, grad_scale=1.0):
+        if self._lr_buf.numel() and \
+                not torch.cuda.is_current_stream_capturing():
+            # buffer mode outside capture: keep the device LR current
+            self._lr_buf.fill_(float(lr))
         self.ext.multi_tensor_sgd(self._t_tensors, self._t_chunks,
-                                  self._nchunks, float(lr), self.momentum,
-                                  self.dampening, self.nesterov,
-                                  float(grad_scale))
+                                  self._nchunks, float(lr), self._lr_buf,
+                                  self.momentum, self.dampening,
+                                  self.nesterov, float(grad_scale))
 
 
 class PackTable:

@@ -63,10 +63,15 @@ struct Chunk {
 // v = mu*v + (g + wd*p); p -= lr * (nesterov ? g + wd*p + mu*v : v)
 // Momentum buffers are zero-initialized host-side, so the first step's
 // v = mu*0 + d_p == d_p matches torch.optim.SGD's clone-on-first-step.
+// lr_ptr: optional device-resident learning rate — lets a hipGraph
+// capture of the step keep a changing LR schedule without re-capture
+// (the host fills the buffer before each replay).
 __global__ __launch_bounds__(kBlock) void multi_tensor_sgd_kernel(
     const Chunk* __restrict__ chunks, int nchunks,
-    const TensorTriple* __restrict__ tensors, float lr, float momentum,
+    const TensorTriple* __restrict__ tensors, float lr,
+    const float* __restrict__ lr_ptr, float momentum,
     float dampening, bool nesterov, float grad_scale) {
+  if (lr_ptr != nullptr) lr = *lr_ptr;
   for (int c = blockIdx.x; c < nchunks; c += gridDim.x) {
     const Chunk ck = chunks[c];
     const TensorTriple t = tensors[ck.tensor_idx];
@@ -321,16 +326,20 @@ std::vector<torch::Tensor> build_sgd_table(
 }
 
 void multi_tensor_sgd(torch::Tensor t_tensors, torch::Tensor t_chunks,
-                      long nchunks, double lr, double momentum,
-                      double dampening, bool nesterov, double grad_scale) {
+                      long nchunks, double lr, torch::Tensor lr_buf,
+                      double momentum, double dampening, bool nesterov,
+                      double grad_scale) {
+  const float* lr_ptr =
+      lr_buf.defined() && lr_buf.numel() > 0
+          ? lr_buf.data_ptr<float>() : nullptr;
   hipLaunchKernelGGL(multi_tensor_sgd_kernel, dim3(grid_for(nchunks)),
                      dim3(kBlock), 0, c10::hip::getCurrentHIPStream().stream(),
                      reinterpret_cast<const Chunk*>(t_chunks.data_ptr()),
                      (int)nchunks,
                      reinterpret_cast<const TensorTriple*>(
                          t_tensors.data_ptr()),
-                     (float)lr, (float)momentum, (float)dampening, nesterov,
-                     (float)grad_scale);
+                     (float)lr, lr_ptr, (float)momentum, (float)dampening,
+                     nesterov, (float)grad_scale);
   CHECK_HIP(hipGetLastError());
 }
 

@@ -87,3 +87,27 @@ def test_replay_loss_finite_and_params_move():
                        for p in t.net.parameters()])
     assert torch.isfinite(torch.tensor(gstep.loss()))
     assert not torch.allclose(before, after)
+
+
+def test_lr_buffer_schedule_across_replays():
+    """The LR schedule must take effect across graph replays through
+    the fused SGD's device LR buffer (no re-capture)."""
+    from mgwfbp_amd.graph_step import GraphedTrainStep
+    t, o = _make('lenet')
+    gstep = GraphedTrainStep(t, o, warmup=2)
+    assert o._fused_sgd is not None
+
+    p0 = torch.cat([p.detach().reshape(-1).clone()
+                    for p in t.net.parameters()])
+    gstep.set_lr(0.0)          # zero LR -> replay must not move params
+    gstep.step()
+    torch.cuda.synchronize()
+    p1 = torch.cat([p.detach().reshape(-1)
+                    for p in t.net.parameters()])
+    assert torch.equal(p0, p1), 'params moved despite lr=0'
+    gstep.set_lr(0.05)
+    gstep.step()
+    torch.cuda.synchronize()
+    p2 = torch.cat([p.detach().reshape(-1)
+                    for p in t.net.parameters()])
+    assert not torch.allclose(p0, p2), 'params frozen with lr>0'
