@@ -684,15 +684,38 @@ def train_with_single(dnn, dataset, data_dir, nworkers, lr, batch_size,
     times = []
     display = 40
     hidden = trainer.net.init_hidden() if dnn == 'lstm' else None
+    want_graph = (os.environ.get('MGX_HIP_GRAPH', '0') == '1'
+                  and torch.cuda.is_available()
+                  and dnn not in ('lstm', 'lstman4')
+                  and nsteps_update == 1)
+    gstep = None
     for epoch in range(max_epochs):
         if trainer.train_iter >= iters_per_epoch * max_epochs:
             break
         for i in range(iters_per_epoch):
             s = time.time()
-            trainer.zero_grad()
-            for _ in range(nsteps_update):
-                _, hidden = trainer.train(1, hidden=hidden)
-            trainer.update_model()
+            if want_graph and gstep is None and trainer.train_iter >= 3:
+                try:
+                    from .graph_step import GraphedTrainStep
+                    gstep = GraphedTrainStep(trainer, trainer.optimizer)
+                    trainer.train_iter += 2   # capture warmup steps
+                except Exception as e:
+                    logger.warning('hipGraph capture failed (%s); '
+                                   'staying eager', e)
+                    want_graph = False
+            if gstep is not None:
+                lr = trainer.adjust_learning_rate(trainer.train_epoch,
+                                                  trainer.optimizer)
+                gstep.set_lr(lr)
+                gstep.step()
+                trainer.train_iter += 1
+                if trainer.train_iter % iters_per_epoch == 0:
+                    trainer.train_epoch += 1
+            else:
+                trainer.zero_grad()
+                for _ in range(nsteps_update):
+                    _, hidden = trainer.train(1, hidden=hidden)
+                trainer.update_model()
             times.append(time.time() - s)
             if i % display == 0 and i > 0:
                 avg = sum(times[-display:]) / min(len(times), display)
