@@ -88,7 +88,7 @@ class BucketingSampler(Sampler):
     is minimal."""
 
     def __init__(self, data_source, batch_size=1):
-        super().__init__(data_source)
+        super().__init__()
         ids = list(range(len(data_source)))
         self.bins = [ids[i:i + batch_size]
                      for i in range(0, len(ids), batch_size)]
@@ -109,7 +109,7 @@ class DistributedBucketingSampler(Sampler):
     """Rank-sharded bucketing sampler (world-size aware)."""
 
     def __init__(self, data_source, batch_size=1, num_replicas=1, rank=0):
-        super().__init__(data_source)
+        super().__init__()
         self.num_replicas = num_replicas
         self.rank = rank
         ids = list(range(len(data_source)))

@@ -9,7 +9,7 @@ from torch.utils.data import Dataset
 
 def _read_words(filename):
     with open(filename, 'r') as f:
-        return f.read().replace('\n', '<eos>').split()
+        return f.read().replace('\n', ' <eos> ').split()
 
 
 def build_vocab(filename):
