@@ -81,7 +81,52 @@ __global__ __launch_bounds__(kBlock) void multi_tensor_sgd_kernel(
     float* __restrict__ m = t.m ? t.m + ck.start : nullptr;
     const float wd = t.wd;
     const long nvec = n & ~3L;  // bases are 16B-aligned; vectorize in 4s
-    for (long i = threadIdx.x * 4L; i < nvec; i += (long)blockDim.x * 4L) {
+    // 2x unrolled main loop: 6-8 independent 16B loads in flight per
+    // thread (single-slot loops leave the kernel latency-bound)
+    const long step = (long)blockDim.x * 4L;
+    long i = threadIdx.x * 4L;
+    for (; i + step < nvec; i += 2 * step) {
+      float4 pv0 = *reinterpret_cast<float4*>(p + i);
+      float4 gv0 = *reinterpret_cast<float4*>(g + i);
+      float4 pv1 = *reinterpret_cast<float4*>(p + i + step);
+      float4 gv1 = *reinterpret_cast<float4*>(g + i + step);
+      float4 mv0, mv1;
+      if (m) {
+        mv0 = *reinterpret_cast<float4*>(m + i);
+        mv1 = *reinterpret_cast<float4*>(m + i + step);
+      }
+#pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        float4& pv = u ? pv1 : pv0;
+        const float4& gv = u ? gv1 : gv0;
+        float4& mv = u ? mv1 : mv0;
+        float dp0 = gv.x * grad_scale + wd * pv.x;
+        float dp1 = gv.y * grad_scale + wd * pv.y;
+        float dp2 = gv.z * grad_scale + wd * pv.z;
+        float dp3 = gv.w * grad_scale + wd * pv.w;
+        if (m) {
+          mv.x = momentum * mv.x + (1.f - dampening) * dp0;
+          mv.y = momentum * mv.y + (1.f - dampening) * dp1;
+          mv.z = momentum * mv.z + (1.f - dampening) * dp2;
+          mv.w = momentum * mv.w + (1.f - dampening) * dp3;
+          if (nesterov) {
+            dp0 += momentum * mv.x; dp1 += momentum * mv.y;
+            dp2 += momentum * mv.z; dp3 += momentum * mv.w;
+          } else {
+            dp0 = mv.x; dp1 = mv.y; dp2 = mv.z; dp3 = mv.w;
+          }
+        }
+        pv.x -= lr * dp0; pv.y -= lr * dp1;
+        pv.z -= lr * dp2; pv.w -= lr * dp3;
+      }
+      if (m) {
+        *reinterpret_cast<float4*>(m + i) = mv0;
+        *reinterpret_cast<float4*>(m + i + step) = mv1;
+      }
+      *reinterpret_cast<float4*>(p + i) = pv0;
+      *reinterpret_cast<float4*>(p + i + step) = pv1;
+    }
+    for (; i < nvec; i += step) {
       float4 pv = *reinterpret_cast<float4*>(p + i);
       float4 gv = *reinterpret_cast<float4*>(g + i);
       float dp0 = gv.x * grad_scale + wd * pv.x;
