@@ -69,7 +69,6 @@ class _DistributedOptimizer(torch.optim.Optimizer):
             'fp32': torch.float32, 'bf16': torch.bfloat16,
             'fp16': torch.float16}[settings.COMM_DTYPE]
         self._allreduce_timers = {}
-        self._update_times = {}
 
         if named_parameters is not None:
             named_parameters = list(named_parameters)
@@ -414,10 +413,10 @@ class _DistributedOptimizer(torch.optim.Optimizer):
 
 
 def DistributedOptimizer(optimizer, named_parameters=None, compression=None,
-                         is_sparse=False, density=1.0, seq_layernames=None,
+                         density=1.0, seq_layernames=None,
                          layerwise_times=None, norm_clip=None, threshold=0,
-                         writer=None, gradient_path=None, alpha=None,
-                         beta=None):
+                         writer=None, gradient_path=None, is_sparse=False,
+                         alpha=None, beta=None):
     """Wrap a torch optimizer with merged-gradient WFBP data parallelism.
 
     Same factory surface as the reference (distributed_optimizer.py:435-471)

@@ -21,7 +21,6 @@ import torch
 
 _kernels_ext = None
 _comm_ext = None
-_warned_cpu = False
 
 
 class ExtensionMissing(RuntimeError):
