@@ -116,6 +116,7 @@ def mgwfbp(dnn, dataset, data_dir, nworkers, lr, batch_size, nsteps_update,
             trainer.train_iter += 1
             if trainer.train_iter % iters_per_epoch == 0:
                 trainer.train_epoch += 1
+            optimizer.train_epoch = trainer.train_epoch
             times.append(time.time() - s)
         else:
             optimizer.zero_grad()
@@ -128,6 +129,7 @@ def mgwfbp(dnn, dataset, data_dir, nworkers, lr, batch_size, nsteps_update,
                 torch.nn.utils.clip_grad_norm_(trainer.net.parameters(),
                                                norm_clip)
             trainer.update_model()
+            optimizer.train_epoch = trainer.train_epoch
             times.append(time.time() - s)
         if i % display == 0 and i > 0 and rank == 0:
             avg = sum(times[-display:]) / min(len(times), display)

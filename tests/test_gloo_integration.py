@@ -114,3 +114,71 @@ def test_bf16_wire_format():
     ref = _reference_dp(world)
     assert torch.allclose(results[0], ref, atol=5e-2, rtol=5e-2), \
         (results[0] - ref).abs().max()
+
+
+def _accum_worker(rank, world, port, q):
+    os.environ['MASTER_ADDR'] = '127.0.0.1'
+    os.environ['MASTER_PORT'] = str(port)
+    os.environ['WORLD_SIZE'] = str(world)
+    os.environ['RANK'] = str(rank)
+    os.environ['MGX_COMM_BACKEND'] = 'gloo'
+    import mgwfbp_amd.comm as comm
+    from mgwfbp_amd import models
+    from mgwfbp_amd.distributed_optimizer import (DistributedOptimizer,
+                                                  broadcast_parameters)
+    comm.init()
+    torch.manual_seed(42 + rank)
+    net = models.LeNet()
+    broadcast_parameters(net.state_dict(), root_rank=0)
+    opt = DistributedOptimizer(
+        torch.optim.SGD(net.parameters(), lr=0.1, momentum=0.9),
+        named_parameters=list(net.named_parameters()), threshold=0)
+    crit = nn.CrossEntropyLoss()
+    # 2 microbatches per step with comm suppressed on the first
+    opt.zero_grad()
+    for j, seed in enumerate((rank * 2, rank * 2 + 1)):
+        opt.local = (j == 0)
+        x, y = _data(seed=seed)
+        crit(net(x), y).backward()
+    opt.local = False
+    opt.step()
+    flat = torch.cat([p.detach().reshape(-1) for p in net.parameters()])
+    q.put((rank, flat.tolist()))
+    comm.shutdown()
+
+
+def test_gradient_accumulation_two_process():
+    """nsteps_update-style accumulation: comm suppressed on microstep 0,
+    accumulated gradient all-reduced on the final microstep."""
+    world = 2
+    ctx = mp.get_context('spawn')
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_accum_worker, args=(r, world, 29631, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(world):
+        rank, flat = q.get()
+        res[rank] = torch.tensor(flat)
+    for p in procs:
+        p.join(180)
+        assert p.exitcode == 0
+    assert torch.equal(res[0], res[1])
+    # sequential reference: average over ranks of SUMMED microbatch grads
+    from mgwfbp_amd import models
+    torch.manual_seed(42)
+    net = models.LeNet()
+    opt = torch.optim.SGD(net.parameters(), lr=0.1, momentum=0.9)
+    crit = nn.CrossEntropyLoss()
+    opt.zero_grad()
+    loss = 0
+    for r in range(world):
+        for seed in (r * 2, r * 2 + 1):
+            x, y = _data(seed=seed)
+            loss = loss + crit(net(x), y) / world
+    loss.backward()
+    opt.step()
+    ref = torch.cat([p.detach().reshape(-1) for p in net.parameters()])
+    assert torch.allclose(res[0], ref, atol=1e-5), \
+        (res[0] - ref).abs().max()
