@@ -37,6 +37,11 @@ def main():
     parser.add_argument('--profile-iters', type=int, default=15,
                         help='layerwise-profiling iterations for the '
                              'solver (mgwfbp arm)')
+    parser.add_argument('--scaling', type=str, default='weak',
+                        choices=['weak', 'strong'],
+                        help='weak: per-GPU batch fixed (BASELINE '
+                             'contract); strong: global batch fixed, '
+                             'split over ranks')
     parser.add_argument('--graph', type=str, default='auto',
                         choices=['auto', 'on', 'off'],
                         help='hipGraph-capture the training step (full '
@@ -88,7 +93,8 @@ def main():
         threshold = int(args.merge.split(':', 1)[1])
 
     trainer = DLTrainer(rank, n_gpus, dist=False,
-                        batch_size=args.batch_size, is_weak_scaling=True,
+                        batch_size=args.batch_size,
+                        is_weak_scaling=(args.scaling == 'weak'),
                         ngpus=1 if torch.cuda.is_available() else 0,
                         data_dir='', dataset=args.dataset, dnn=args.model,
                         lr=0.01, nworkers=n_gpus, prefix='bench',
@@ -183,7 +189,7 @@ def main():
         elapsed = float(t2.cpu()[0])
 
     ms_per_step = elapsed / args.steps * 1e3
-    global_batch = args.batch_size * n_gpus
+    global_batch = trainer.batch_size * n_gpus
     images_per_sec = global_batch * args.steps / elapsed
 
     if rank == 0:
@@ -198,14 +204,14 @@ def main():
             'warmup': args.warmup,
             'ms_per_step': ms_per_step,
             'higher_is_better': True,
-            'scaling': 'weak',
+            'scaling': args.scaling,
             'vs_baseline': None,
             'dtype': args.dtype,
             'data': 'synthetic',
             'config': {
                 'model': args.model,
                 'global_batch': global_batch,
-                'per_gpu_batch': args.batch_size,
+                'per_gpu_batch': trainer.batch_size,
                 'image_size': 224 if args.dataset == 'imagenet' else 32,
                 'parallelism': 'dp%d' % n_gpus,
                 'merge': args.merge,
