@@ -208,6 +208,13 @@ class DLTrainer:
 
         self.train_epoch = 0
         self.train_iter = 0
+        # device-side loss accumulation: loss.item() every step forces a
+        # full host sync per iteration (the reference's natural sync
+        # point) — on GPU we read the loss back only at
+        # display boundaries so step boundaries pipeline
+        self._async_loss = (self.is_cuda and
+                            os.environ.get('MGX_ASYNC_LOSS', '1') == '1')
+        self._loss_accum = None
         self.avg_loss_per_epoch = 0.0
         self.accuracy = 0.0
         self.loss = 0.0
@@ -481,6 +488,11 @@ class DLTrainer:
             logger.info(
                 'Time per iteration including communication: %f, Speed: '
                 '%f images/s', avg, throughput)
+            if self._async_loss and self._loss_accum is not None:
+                mean_loss = float(self._loss_accum.item()) / self.display
+                self._loss_accum.zero_()
+                logger.info('  mean loss (last %d iters): %.4f',
+                            self.display, mean_loss)
             logger.info('  phase times (last %d iters): io %.4f fwd %.4f '
                         'bwd %.4f', n, self.io_time / self.display,
                         self.forward_time / self.display,
