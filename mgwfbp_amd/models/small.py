@@ -64,11 +64,12 @@ class FCN5Net(nn.Module):
 
 
 class LinearRegression(nn.Module):
-    """1-D linear regression (reference fcn.py:28-35, dnn='lr')."""
+    """Linear model (reference fcn.py:28-35, dnn='lr'; the reference
+    config pairs it with MNIST, so inputs are flattened)."""
 
-    def __init__(self, in_features=1, out_features=1):
+    def __init__(self, in_features=784, out_features=10):
         super().__init__()
         self.linear = nn.Linear(in_features, out_features)
 
     def forward(self, x):
-        return self.linear(x)
+        return self.linear(x.flatten(1))

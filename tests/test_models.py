@@ -41,7 +41,7 @@ def test_small_models():
     assert models.LeNet()(torch.randn(2, 3, 32, 32)).shape == (2, 10)
     assert models.MnistNet()(torch.randn(2, 1, 28, 28)).shape == (2, 10)
     assert models.FCN5Net()(torch.randn(2, 1, 28, 28)).shape == (2, 10)
-    assert models.LinearRegression()(torch.randn(2, 1)).shape == (2, 1)
+    assert models.LinearRegression()(torch.randn(2, 1, 28, 28)).shape == (2, 10)
 
 
 def test_ptb_lstm():
