@@ -224,14 +224,14 @@ constexpr int kFinC = 32;                  // channels per finalize block
 constexpr int kFinG = kBlock / kFinC;      // row-subsets per block
 
 __device__ inline void finalize_sums(const float* __restrict__ partial,
-                                     long nblk, long C, long& c,
+                                     long b0, long b1, long C, long& c,
                                      bool& leader, float& s, float& q) {
   const int cl = threadIdx.x % kFinC;
   const int g = threadIdx.x / kFinC;
   c = (long)blockIdx.x * kFinC + cl;
   s = 0.f; q = 0.f;
   if (c < C) {
-    for (long b = g; b < nblk; b += kFinG) {
+    for (long b = b0 + g; b < b1; b += kFinG) {
       s += partial[b * 2 * C + c];
       q += partial[b * 2 * C + C + c];
     }
@@ -249,13 +249,29 @@ __device__ inline void finalize_sums(const float* __restrict__ partial,
   }
 }
 
+// Two-level tree for large nblk: stage1 splits the block range
+// gridDim.y ways (PMC: a single-level finalize at nblk=1024, C=64 is
+// 33 us latency-bound — 128 strided loads per thread on 2 blocks).
+__global__ __launch_bounds__(kBlock) void bn_finalize_stage1_kernel(
+    const float* __restrict__ partial, long nblk, long C,
+    float* __restrict__ stage) {
+  const long chunk = (nblk + gridDim.y - 1) / gridDim.y;
+  const long b0 = (long)blockIdx.y * chunk;
+  const long b1 = b0 + chunk < nblk ? b0 + chunk : nblk;
+  long c; bool leader; float s, q;
+  finalize_sums(partial, b0, b1, C, c, leader, s, q);
+  if (!leader) return;
+  stage[(long)blockIdx.y * 2 * C + c] = s;
+  stage[(long)blockIdx.y * 2 * C + C + c] = q;
+}
+
 __global__ __launch_bounds__(kBlock) void bn_fwd_finalize_kernel(
     const float* __restrict__ partial, long nblk, long C, float M,
     float eps, float momentum, float* __restrict__ mean,
     float* __restrict__ invstd, float* __restrict__ running_mean,
     float* __restrict__ running_var) {
   long c; bool leader; float s, q;
-  finalize_sums(partial, nblk, C, c, leader, s, q);
+  finalize_sums(partial, 0, nblk, C, c, leader, s, q);
   if (!leader) return;
   const float m = s / M;
   float var = q / M - m * m;
@@ -405,7 +421,7 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_finalize_kernel(
     const float* __restrict__ partial, long nblk, long C,
     float* __restrict__ dbeta, float* __restrict__ dgamma) {
   long c; bool leader; float sd, sx;
-  finalize_sums(partial, nblk, C, c, leader, sd, sx);
+  finalize_sums(partial, 0, nblk, C, c, leader, sd, sx);
   if (!leader) return;
   dbeta[c] = sd;
   dgamma[c] = sx;
@@ -512,6 +528,22 @@ int grid_elem(const Geometry& g) {
   return (int)(nblk < 1 ? 1 : nblk);
 }
 
+constexpr int kFinSplit = 8;
+
+// Returns the (buffer, count) the final finalize kernel should read:
+// for large nblk, runs the stage1 tree first.
+std::pair<float*, long> finalize_tree(torch::Tensor& partial, int rblocks,
+                                      long C, hipStream_t stream) {
+  float* base = partial.data_ptr<float>();
+  if (rblocks <= 64) return {base, (long)rblocks};
+  float* stage = base + (long)rblocks * 2 * C;
+  hipLaunchKernelGGL(bn_finalize_stage1_kernel,
+                     dim3((C + kFinC - 1) / kFinC, kFinSplit),
+                     dim3(kBlock), 0, stream, base, (long)rblocks, C,
+                     stage);
+  return {stage, (long)kFinSplit};
+}
+
 #define DISPATCH_DT(scalar_type, ...)                                        \
   switch (scalar_type) {                                                     \
     case at::kFloat: {                                                       \
@@ -535,7 +567,8 @@ std::vector<torch::Tensor> bn_fwd_train(
   auto opts = torch::TensorOptions().dtype(torch::kFloat).device(x.device());
   const int rblocks = grid_reduce(g);
   const int eblocks = grid_elem(g);
-  auto partial = torch::empty({(long)rblocks * 2 * g.C}, opts);
+  auto partial = torch::empty(
+      {(long)(rblocks + (rblocks > 64 ? kFinSplit : 0)) * 2 * g.C}, opts);
   auto mean = torch::empty({g.C}, opts);
   auto invstd = torch::empty({g.C}, opts);
   auto y = torch::empty_like(x);
@@ -545,9 +578,10 @@ std::vector<torch::Tensor> bn_fwd_train(
                        reinterpret_cast<const dt*>(x.data_ptr()), g.rows,
                        g.C, partial.data_ptr<float>());
   });
+  auto fin = finalize_tree(partial, rblocks, g.C, stream);
   hipLaunchKernelGGL(bn_fwd_finalize_kernel,
                      dim3((g.C + kFinC - 1) / kFinC), dim3(kBlock), 0,
-                     stream, partial.data_ptr<float>(), rblocks, g.C,
+                     stream, fin.first, fin.second, g.C,
                      (float)g.rows, (float)eps, (float)momentum,
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      running_mean.defined()
@@ -603,7 +637,8 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   auto opts = torch::TensorOptions().dtype(torch::kFloat).device(x.device());
   const int rblocks = grid_reduce(g);
   const int eblocks = grid_elem(g);
-  auto partial = torch::empty({(long)rblocks * 2 * g.C}, opts);
+  auto partial = torch::empty(
+      {(long)(rblocks + (rblocks > 64 ? kFinSplit : 0)) * 2 * g.C}, opts);
   auto dbeta = torch::empty({g.C}, opts);
   auto dgamma = torch::empty({g.C}, opts);
   auto dx = torch::empty_like(x);
@@ -628,9 +663,10 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                          invstd.data_ptr<float>(), gamma_p, beta_p,
                          partial.data_ptr<float>());
   });
+  auto finb = finalize_tree(partial, rblocks, g.C, stream);
   hipLaunchKernelGGL(bn_bwd_finalize_kernel,
                      dim3((g.C + kFinC - 1) / kFinC), dim3(kBlock), 0,
-                     stream, partial.data_ptr<float>(), rblocks, g.C,
+                     stream, finb.first, finb.second, g.C,
                      dbeta.data_ptr<float>(), dgamma.data_ptr<float>());
   DISPATCH_DT(x.scalar_type(), {
     if (relu)
