@@ -57,7 +57,7 @@ def time_raw(x, iters=20):
     for _ in range(3):
         y, mean, invstd = ext.bn_fwd_train(x, w, b, rm, rv, 0.1, 1e-5,
                                            False)
-        ext.bn_bwd(dy, x, mean, invstd, w)
+        ext.bn_bwd(dy, x, mean, invstd, w, b, False)
     torch.cuda.synchronize()
     s = torch.cuda.Event(True)
     m = torch.cuda.Event(True)
@@ -68,7 +68,7 @@ def time_raw(x, iters=20):
         y, mean, invstd = ext.bn_fwd_train(x, w, b, rm, rv, 0.1, 1e-5,
                                            False)
         m.record()
-        ext.bn_bwd(dy, x, mean, invstd, w)
+        ext.bn_bwd(dy, x, mean, invstd, w, b, False)
         e.record()
         torch.cuda.synchronize()
         tf += s.elapsed_time(m)
