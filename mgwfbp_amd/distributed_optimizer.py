@@ -125,14 +125,26 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         prof = CommunicationProfiler(comm.allreduce_async_, comm.synchronize)
         sizes, times = prof.benchmark(num_iters=10)
         nbytes = torch.tensor([], dtype=self._comm_dtype).element_size()
-        a, b = solver.fit_alpha_beta([s * nbytes for s in sizes], times)
+        sweep_bytes = [s * nbytes for s in sizes]
+        a, b = solver.fit_alpha_beta(sweep_bytes, times)
         t = torch.tensor([a, b], dtype=torch.float64)
         if torch.cuda.is_available():
             t = t.cuda()   # RCCL core broadcasts device tensors
+        # broadcast the full measured table too (rank 0 authoritative)
+        # so the solver can interpolate instead of relying on the fit
+        # (the reference's never-populated size_commtime_dict hook)
+        tab = torch.tensor([sweep_bytes, times], dtype=torch.float64)
+        if torch.cuda.is_available():
+            tab = tab.cuda()
         comm.broadcast(t, root_rank=0)
+        comm.broadcast(tab, root_rank=0)
         self.alpha, self.beta = float(t[0].item()), float(t[1].item())
-        logger.info('[rank %d] fitted allreduce model t = %.3e + %.3e * bytes',
-                    comm.rank(), self.alpha, self.beta)
+        tab = tab.cpu()
+        self.size_commtime_dict = ([float(x) for x in tab[0]],
+                                   [float(x) for x in tab[1]])
+        logger.info('[rank %d] fitted allreduce model t = %.3e + %.3e * bytes'
+                    ' (+%d-point measured table)',
+                    comm.rank(), self.alpha, self.beta, tab.shape[1])
 
     # ------------------------------------------------------------------
     # merge groups + flat buffers (+ grad views)
@@ -148,7 +160,7 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                     settings.CONNECTION, max(comm.size(), 2))
             groups, gmap, stats = solver.generate_groups_mgwfbp(
                 keys, self._layerwise_times, sizes, self.alpha, self.beta,
-                nbytes)
+                nbytes, size_commtime=self.size_commtime_dict)
             if comm.rank() == 0:
                 logger.info(
                     'MG-WFBP solver: %d layers -> %d groups; predicted '

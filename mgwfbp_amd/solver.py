@@ -76,6 +76,35 @@ def predict_allreduce_time(alpha: float, beta: float, size_bytes: float) -> floa
     return alpha + beta * size_bytes
 
 
+def predict_from_table(table_sizes, table_times, size_bytes):
+    """Piecewise-linear interpolation over a MEASURED (bytes -> seconds)
+    all-reduce sweep — the reference's ``size_commtime_dict`` hook
+    (reference distributed_optimizer.py:197-199, :210-212), which its
+    entry points never populate; here the online sweep can feed it so
+    non-linear RCCL protocol switches (LL/LL128/simple) are captured
+    instead of forced through one alpha+beta line.
+
+    table_sizes must be ascending. Extrapolates linearly from the last
+    segment above the table, and scales proportionally below it.
+    """
+    if size_bytes <= 0:
+        return 0.0
+    n = len(table_sizes)
+    if n == 0:
+        raise ValueError('empty comm table')
+    if n == 1 or size_bytes <= table_sizes[0]:
+        return table_times[0] * size_bytes / table_sizes[0]             if size_bytes < table_sizes[0] else table_times[0]
+    import bisect
+    i = bisect.bisect_left(table_sizes, size_bytes)
+    if i >= n:
+        i = n - 1
+    lo, hi = i - 1, i
+    if table_sizes[hi] == table_sizes[lo]:
+        return table_times[hi]
+    f = (size_bytes - table_sizes[lo]) / (table_sizes[hi] - table_sizes[lo])
+    return table_times[lo] + f * (table_times[hi] - table_times[lo])
+
+
 def _comm_start_times(tc: List[float], tb: Sequence[float],
                       taob: Sequence[float], L: int) -> List[float]:
     """Earliest all-reduce start per layer given channel serialization.
@@ -99,6 +128,7 @@ def generate_groups_mgwfbp(
     alpha: float,
     beta: float,
     nbytes: int = 4,
+    size_commtime: 'Tuple[Sequence[float], Sequence[float]]' = None,
 ) -> Tuple[List[List[str]], Dict[str, int], Dict[str, float]]:
     """Solve the optimal merged-gradient grouping.
 
@@ -120,7 +150,17 @@ def generate_groups_mgwfbp(
 
     tb = list(layerwise_times)
     p = [int(s) for s in sizes]          # merged element counts (mutated)
-    tc = [predict_allreduce_time(alpha, beta, s * nbytes) for s in p]
+
+    if size_commtime is not None:
+        t_sizes, t_times = size_commtime
+
+        def comm_cost(size_bytes):
+            return predict_from_table(t_sizes, t_times, size_bytes)
+    else:
+        def comm_cost(size_bytes):
+            return predict_allreduce_time(alpha, beta, size_bytes)
+
+    tc = [comm_cost(s * nbytes) for s in p]
     # Gradient-ready offsets: taob[L-1] = 0 (backward starts at the last
     # layer); taob[l] = taob[l+1] + tb[l+1].
     taob = [0.0] * L
@@ -135,7 +175,7 @@ def generate_groups_mgwfbp(
         p[l - 1] += p[l]
         p[l] = 0
         tc[l] = 0.0
-        tc[l - 1] = predict_allreduce_time(alpha, beta, p[l - 1] * nbytes)
+        tc[l - 1] = comm_cost(p[l - 1] * nbytes)
 
     groups: List[List[str]] = []
     group: List[str] = []

@@ -152,3 +152,42 @@ class TestAlphaBetaFit:
         # unknown world size falls back to nearest
         a2, _ = solver.lookup_alpha_beta('xgmi', 3)
         assert a2 > 0
+
+
+class TestMeasuredCommTable:
+    def test_interp_matches_linear_table(self):
+        sizes = [1000.0 * i for i in range(1, 20)]
+        times = [2e-5 + 1e-11 * s for s in sizes]
+        for q in (1500.0, 7777.0, 19000.0, 50000.0):
+            got = solver.predict_from_table(sizes, times, q)
+            if q <= sizes[-1]:
+                assert abs(got - (2e-5 + 1e-11 * q)) < 1e-12
+        # below the table scales proportionally, zero -> zero
+        assert solver.predict_from_table(sizes, times, 0) == 0.0
+        assert solver.predict_from_table(sizes, times, 500.0) == \
+            times[0] * 0.5
+
+    def test_solver_with_table_equals_linear_when_table_linear(self):
+        names = ['l%d' % i for i in range(12)]
+        tb = [5e-5] * 12
+        szs = [10000 + 1000 * i for i in range(12)]
+        alpha, beta = 3e-5, 2e-10
+        tsizes = [4.0 * 2 ** k for k in range(8, 30)]
+        ttimes = [alpha + beta * s for s in tsizes]
+        g1, _, s1 = solver.generate_groups_mgwfbp(names, tb, szs, alpha,
+                                                  beta, 4)
+        g2, _, s2 = solver.generate_groups_mgwfbp(
+            names, tb, szs, alpha, beta, 4,
+            size_commtime=(tsizes, ttimes))
+        assert g1 == g2
+        assert abs(s1['predicted_total_time']
+                   - s2['predicted_total_time']) < 1e-9
+
+    def test_nonlinear_table_respected(self):
+        # a protocol-switch-like jump in the table must show in tc
+        sizes = [1e3, 1e6, 1e6 + 1, 1e8]
+        times = [1e-5, 2e-5, 2e-4, 1e-3]
+        below = solver.predict_from_table(sizes, times, 5e5)
+        above = solver.predict_from_table(sizes, times, 2e6)
+        assert below < 2e-5
+        assert above > 2e-4
