@@ -121,6 +121,15 @@ class TorchDistBackend:
         work = dist.broadcast(tensor, src=root, async_op=True)
         return _TorchWorkHandle(work, tensor)
 
+    def allgather_async(self, tensor):
+        """All-gather equal-shaped tensors; returns (handle, outputs)
+        where outputs is a list of per-rank tensors valid after wait()."""
+        if self.size() == 1:
+            return _NoopHandle(tensor), [tensor]
+        outs = [torch.empty_like(tensor) for _ in range(self.size())]
+        work = dist.all_gather(outs, tensor, async_op=True)
+        return _TorchWorkHandle(work, tensor), outs
+
     def broadcast(self, tensor, root):
         if self.size() > 1:
             dist.broadcast(tensor, src=root)
@@ -214,6 +223,16 @@ class RcclCoreBackend:
         hid = self._core.broadcast_async(
             tensor, root, torch.cuda.current_stream().cuda_stream)
         return _RcclCoreHandle(self._core, hid, tensor)
+
+    def allgather_async(self, tensor):
+        if self._size == 1:
+            return _NoopHandle(tensor), [tensor]
+        recv = torch.empty((self._size,) + tuple(tensor.shape),
+                           dtype=tensor.dtype, device=tensor.device)
+        hid = self._core.allgather_async(
+            tensor, recv, torch.cuda.current_stream().cuda_stream)
+        return (_RcclCoreHandle(self._core, hid, recv),
+                [recv[r] for r in range(self._size)])
 
     def broadcast(self, tensor, root):
         h = self.broadcast_async(tensor, root)
@@ -316,6 +335,14 @@ def broadcast(tensor, root_rank=0, name=None):
 
 def broadcast_async_(tensor, root_rank=0, name=None):
     return _get().broadcast_async(tensor, root_rank)
+
+
+def allgather_async_(tensor, name=None):
+    """Async all-gather; returns (handle, list-of-per-rank-tensors) —
+    the reference imported Horovod's allgather_async without using it
+    (reference distributed_optimizer.py:22); here it carries the top-k
+    sparse gradient exchange."""
+    return _get().allgather_async(tensor)
 
 
 def barrier():

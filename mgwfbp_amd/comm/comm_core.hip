@@ -149,6 +149,30 @@ long broadcast_async(torch::Tensor tensor, int root,
   return enqueue_collective(tensor, caller_stream, false, true, root);
 }
 
+long allgather_async(torch::Tensor send, torch::Tensor recv,
+                     uintptr_t caller_stream) {
+  TORCH_CHECK(g_core.initialized, "comm core not initialized");
+  TORCH_CHECK(send.is_cuda() && send.is_contiguous() && recv.is_cuda()
+                  && recv.is_contiguous(),
+              "allgather needs contiguous GPU tensors");
+  TORCH_CHECK(recv.numel() == send.numel() * g_core.size,
+              "recv must be world_size x send");
+  hipStream_t cstream = reinterpret_cast<hipStream_t>(caller_stream);
+  std::lock_guard<std::mutex> lock(g_core.mu);
+  hipEvent_t ready = g_core.get_event();
+  CHECK_HIP(hipEventRecord(ready, cstream));
+  CHECK_HIP(hipStreamWaitEvent(g_core.comm_stream, ready, 0));
+  g_core.put_event(ready);
+  CHECK_NCCL(ncclAllGather(send.data_ptr(), recv.data_ptr(), send.numel(),
+                           nccl_dtype(send.scalar_type()), g_core.comm,
+                           g_core.comm_stream));
+  hipEvent_t done = g_core.get_event();
+  CHECK_HIP(hipEventRecord(done, g_core.comm_stream));
+  long hid = g_core.next_handle++;
+  g_core.handles[hid] = done;
+  return hid;
+}
+
 void wait_handle(long hid, uintptr_t caller_stream) {
   std::lock_guard<std::mutex> lock(g_core.mu);
   auto it = g_core.handles.find(hid);
@@ -193,6 +217,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Async in-place all-reduce on the comm stream; returns handle");
   m.def("broadcast_async", &broadcast_async,
         "Async in-place broadcast on the comm stream; returns handle");
+  m.def("allgather_async", &allgather_async,
+        "Async all-gather send->recv on the comm stream; returns handle");
   m.def("wait_handle", &wait_handle,
         "Caller stream waits the collective's hipEvent (device-side)");
   m.def("wait_handle_host", &wait_handle_host, "Host-blocking wait");

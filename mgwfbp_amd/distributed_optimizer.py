@@ -49,9 +49,12 @@ class _DistributedOptimizer(torch.optim.Optimizer):
     def __init__(self, params, named_parameters, compression=None,
                  is_sparse=False, seq_layernames=None, layerwise_times=None,
                  norm_clip=None, threshold=0, writer=None, gradient_path=None,
-                 alpha=None, beta=None):
+                 alpha=None, beta=None, density=1.0):
         super(self.__class__, self).__init__(params)
         self._compression = compression
+        self._density = density
+        self._is_sparse = bool(is_sparse) or density < 1.0
+        self._group_residuals = {}   # group_key -> error-feedback buffer
         self._profiling = False
         self._seq_layernames = list(seq_layernames) if seq_layernames else None
         self._layerwise_times = (list(layerwise_times)
@@ -329,6 +332,11 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         key = self._group_keys[gi]
         flat = self._merged_parameters[key]
         name = key if len(key) <= 100 else key[0:50] + '...' + key[50:100]
+        if self._is_sparse and comm.size() > 1:
+            handle = self._sparse_allgather_async(key, flat)
+            assert key not in self._handles
+            self._handles[key] = handle
+            return
         if self._comm_dtype == torch.float32:
             handle = comm.allreduce_async_(flat, average=True, name=name)
         else:
@@ -348,6 +356,31 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                 torch.cuda.synchronize()
         self._handles[key] = handle
 
+    def _sparse_allgather_async(self, key, flat):
+        """Top-k sparse exchange for one merge group.
+
+        Residual-corrected gradient -> top-k (values, indices) ->
+        all-gather (equal k per rank, dense-shaped collective) ->
+        SparseGroupHandle rebuilds the averaged gradient on wait().
+        Makes the reference's density/compressor scaffolding functional
+        (no working compressor exists there).
+        """
+        from .compression import TopKCompressor
+        residual = self._group_residuals.get(key)
+        if residual is None:
+            residual = torch.zeros_like(flat)
+            self._group_residuals[key] = residual
+        flat.add_(residual)                      # carry-in error feedback
+        (values, indices), numel = TopKCompressor.compress(
+            flat, density=self._density)
+        # carry-out: everything NOT sent becomes the next residual
+        residual.copy_(flat)
+        residual[indices] = 0
+        h_v, out_v = comm.allgather_async_(values)
+        h_i, out_i = comm.allgather_async_(indices)
+        return _SparseGroupHandle(flat, h_v, h_i, out_v, out_i,
+                                  comm.size())
+
     # ------------------------------------------------------------------
     # synchronize + step
     # ------------------------------------------------------------------
@@ -355,7 +388,9 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         for key, handle in self._handles.items():
             stime = time.time() if self._profiling else 0.0
             handle.wait()
-            if self._comm_dtype != torch.float32:
+            if isinstance(handle, _SparseGroupHandle):
+                pass        # wait() rebuilt the averaged gradient
+            elif self._comm_dtype != torch.float32:
                 flat = self._merged_parameters[key]
                 cbuf = self._group_comm_buffers[key]
                 if key in self._group_pack_tables:
@@ -424,6 +459,33 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         self._hook_handles = []
 
 
+class _SparseGroupHandle:
+    """Waits the two all-gathers and rebuilds the group's averaged
+    gradient: scatter-add every rank's (values, indices) then / P."""
+
+    def __init__(self, flat, h_values, h_indices, out_values, out_indices,
+                 world):
+        self._flat = flat
+        self._hv = h_values
+        self._hi = h_indices
+        self._values = out_values
+        self._indices = out_indices
+        self._world = world
+        self._done = False
+
+    def wait(self):
+        if self._done:
+            return self._flat
+        self._hv.wait()
+        self._hi.wait()
+        self._flat.zero_()
+        for v, i in zip(self._values, self._indices):
+            self._flat.scatter_add_(0, i, v)
+        self._flat.div_(self._world)
+        self._done = True
+        return self._flat
+
+
 def DistributedOptimizer(optimizer, named_parameters=None, compression=None,
                          density=1.0, seq_layernames=None,
                          layerwise_times=None, norm_clip=None, threshold=0,
@@ -443,7 +505,8 @@ def DistributedOptimizer(optimizer, named_parameters=None, compression=None,
                seq_layernames=seq_layernames,
                layerwise_times=layerwise_times, norm_clip=norm_clip,
                threshold=threshold, writer=writer,
-               gradient_path=gradient_path, alpha=alpha, beta=beta)
+               gradient_path=gradient_path, alpha=alpha, beta=beta,
+               density=density)
 
 
 def broadcast_parameters(params, root_rank=0):

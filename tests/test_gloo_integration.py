@@ -182,3 +182,62 @@ def test_gradient_accumulation_two_process():
     ref = torch.cat([p.detach().reshape(-1) for p in net.parameters()])
     assert torch.allclose(res[0], ref, atol=1e-5), \
         (res[0] - ref).abs().max()
+
+
+def _sparse_worker(rank, world, port, q):
+    os.environ['MASTER_ADDR'] = '127.0.0.1'
+    os.environ['MASTER_PORT'] = str(port)
+    os.environ['WORLD_SIZE'] = str(world)
+    os.environ['RANK'] = str(rank)
+    os.environ['MGX_COMM_BACKEND'] = 'gloo'
+    import mgwfbp_amd.comm as comm
+    from mgwfbp_amd import models
+    from mgwfbp_amd.distributed_optimizer import (DistributedOptimizer,
+                                                  broadcast_parameters)
+    comm.init()
+    torch.manual_seed(7 + rank)
+    net = models.LeNet()
+    broadcast_parameters(net.state_dict(), root_rank=0)
+    opt = DistributedOptimizer(
+        torch.optim.SGD(net.parameters(), lr=0.01, momentum=0.9),
+        named_parameters=list(net.named_parameters()), threshold=0,
+        density=0.25)
+    crit = nn.CrossEntropyLoss()
+    # fixed memorization set (same on both ranks for convergence check)
+    g = torch.Generator().manual_seed(11)
+    x = torch.randn(16, 3, 32, 32, generator=g)
+    y = torch.randint(0, 10, (16,), generator=g)
+    first = last = None
+    for step in range(120):
+        opt.zero_grad()
+        loss = crit(net(x), y)
+        loss.backward()
+        opt.step()
+        if first is None:
+            first = loss.item()
+        last = loss.item()
+    flat = torch.cat([p.detach().reshape(-1) for p in net.parameters()])
+    q.put((rank, flat.tolist(), first, last))
+    comm.shutdown()
+
+
+def test_topk_sparse_training_converges_and_ranks_agree():
+    """density=0.25 top-k with error feedback: ranks stay bit-identical
+    and the model still memorizes a fixed batch."""
+    world = 2
+    ctx = mp.get_context('spawn')
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_sparse_worker, args=(r, world, 29641, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(world):
+        rank, flat, first, last = q.get()
+        res[rank] = (torch.tensor(flat), first, last)
+    for p in procs:
+        p.join(300)
+        assert p.exitcode == 0
+    assert torch.equal(res[0][0], res[1][0])
+    first, last = res[0][1], res[0][2]
+    assert last < 0.3 * first, (first, last)
