@@ -37,6 +37,9 @@ def main():
     parser.add_argument('--profile-iters', type=int, default=15,
                         help='layerwise-profiling iterations for the '
                              'solver (mgwfbp arm)')
+    parser.add_argument('--density', type=float, default=1.0,
+                        help='<1 enables top-k sparse gradient exchange '
+                             'with error feedback')
     parser.add_argument('--scaling', type=str, default='weak',
                         choices=['weak', 'strong'],
                         help='weak: per-GPU batch fixed (BASELINE '
@@ -117,7 +120,7 @@ def main():
         trainer.optimizer,
         named_parameters=list(trainer.net.named_parameters()),
         seq_layernames=seq_layernames, layerwise_times=layerwise_times,
-        threshold=threshold)
+        threshold=threshold, density=args.density)
     trainer.update_optimizer(optimizer)
     if comm.size() > 1:
         broadcast_parameters(trainer.net.state_dict(), root_rank=0)
@@ -216,6 +219,7 @@ def main():
                 'merge': args.merge,
                 'comm_backend': comm.backend_name(),
                 'hip_graph': graphed,
+                'density': args.density,
             },
         }
         print(json.dumps(result))
