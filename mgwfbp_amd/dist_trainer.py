@@ -73,9 +73,9 @@ def mgwfbp(dnn, dataset, data_dir, nworkers, lr, batch_size, nsteps_update,
     optimizer = DistributedOptimizer(
         trainer.optimizer,
         named_parameters=list(trainer.net.named_parameters()),
-        compression=compression, is_sparse=is_sparse,
+        compression=compression, is_sparse=is_sparse, density=density,
         seq_layernames=seq_layernames, layerwise_times=layerwise_times,
-        norm_clip=None, threshold=threshold, writer=writer)
+        norm_clip=norm_clip, threshold=threshold, writer=writer)
     trainer.update_optimizer(optimizer)
 
     broadcast_parameters(trainer.net.state_dict(), root_rank=0)
@@ -123,9 +123,13 @@ def mgwfbp(dnn, dataset, data_dir, nworkers, lr, batch_size, nsteps_update,
             for j in range(nsteps_update):
                 optimizer.local = (j < nsteps_update - 1)
                 _, hidden = trainer.train(1, hidden=hidden)
-            if dnn in ('lstm', 'lstman4'):
-                # external clipping path (reference dist_trainer.py:89-94)
-                optimizer.synchronize()
+            if dnn in ('lstm', 'lstman4') and comm.size() <= 1:
+                # single-process: no hooks fire, so the optimizer's
+                # internal device-side per-group clip never runs — clip
+                # externally (reference dist_trainer.py:89-94). At
+                # world>1 norm_clip is honored inside synchronize()
+                # (device-side, no .item(); the reference factory
+                # dropped the arg — SURVEY.md §7.5).
                 torch.nn.utils.clip_grad_norm_(trainer.net.parameters(),
                                                norm_clip)
             trainer.update_model()

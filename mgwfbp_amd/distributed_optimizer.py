@@ -65,6 +65,7 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         self._gradient_path = gradient_path
         self.alpha = alpha
         self.beta = beta
+        self.alpha_host = settings.ALPHA_HOST
         self.train_epoch = 0
         self.train_iter = 0
         self.local = False
@@ -126,11 +127,19 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         logger.info('Benchmarking all-reduce alpha/beta over %s...',
                     settings.CONNECTION)
         prof = CommunicationProfiler(comm.allreduce_async_, comm.synchronize)
-        sizes, times = prof.benchmark(num_iters=10)
+        # sweep in the WIRE dtype so the fitted beta and the measured
+        # table attribute times to the byte counts that actually move
+        sizes, times = prof.benchmark(num_iters=10,
+                                      dtype=self._comm_dtype)
         nbytes = torch.tensor([], dtype=self._comm_dtype).element_size()
         sweep_bytes = [s * nbytes for s in sizes]
         a, b = solver.fit_alpha_beta(sweep_bytes, times)
-        t = torch.tensor([a, b], dtype=torch.float64)
+        # Per-collective HOST cost (async-enqueue launch-to-launch): paid
+        # once per group regardless of device-side overlap — exactly what
+        # merging amortizes at xGMI latencies. Folded into the solver's
+        # per-call constant.
+        a_host = prof.benchmark_host_overhead(dtype=self._comm_dtype)
+        t = torch.tensor([a, b, a_host], dtype=torch.float64)
         if torch.cuda.is_available():
             t = t.cuda()   # RCCL core broadcasts device tensors
         # broadcast the full measured table too (rank 0 authoritative)
@@ -142,12 +151,14 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         comm.broadcast(t, root_rank=0)
         comm.broadcast(tab, root_rank=0)
         self.alpha, self.beta = float(t[0].item()), float(t[1].item())
+        self.alpha_host = float(t[2].item())
         tab = tab.cpu()
         self.size_commtime_dict = ([float(x) for x in tab[0]],
                                    [float(x) for x in tab[1]])
         logger.info('[rank %d] fitted allreduce model t = %.3e + %.3e * bytes'
-                    ' (+%d-point measured table)',
-                    comm.rank(), self.alpha, self.beta, tab.shape[1])
+                    ' + host %.3e/call (+%d-point measured table)',
+                    comm.rank(), self.alpha, self.beta, self.alpha_host,
+                    tab.shape[1])
 
     # ------------------------------------------------------------------
     # merge groups + flat buffers (+ grad views)
@@ -163,7 +174,8 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                     settings.CONNECTION, max(comm.size(), 2))
             groups, gmap, stats = solver.generate_groups_mgwfbp(
                 keys, self._layerwise_times, sizes, self.alpha, self.beta,
-                nbytes, size_commtime=self.size_commtime_dict)
+                nbytes, size_commtime=self.size_commtime_dict,
+                alpha_host=self.alpha_host)
             if comm.rank() == 0:
                 logger.info(
                     'MG-WFBP solver: %d layers -> %d groups; predicted '
@@ -427,19 +439,23 @@ class _DistributedOptimizer(torch.optim.Optimizer):
 
     def _clip_merged(self, flat):
         # reference :380-389: per-merged-tensor L2 clip at
-        # sqrt(1/P)*norm_clip
+        # sqrt(1/P)*norm_clip — device-side here (l2norm_sq_flat +
+        # clip_scale kernels; the reference host-syncs via .item() per
+        # merged tensor)
+        from . import kernels as _k
         norm_clip = (1.0 / comm.size()) ** 0.5 * self._norm_clip
-        total_norm = flat.norm(2).item()
-        clip_coef = norm_clip / (total_norm + 1e-6)
-        if clip_coef < 1:
-            flat.mul_(clip_coef)
+        _k.l2norm_clip_(flat, norm_clip)
 
     def step(self, closure=None):
         if not self.local and comm.size() > 1:
             self.synchronize()
         if self._fused_sgd is not None and closure is None:
-            self._fused_sgd.step(lr=self.param_groups[0]['lr'])
-            return None
+            lr = self.param_groups[0]['lr']
+            if all(g['lr'] == lr for g in self.param_groups):
+                self._fused_sgd.step(lr=lr)
+                return None
+            # per-group LRs diverged (user schedule): the single-launch
+            # fused kernel carries one LR — use the torch loop instead
         return super(self.__class__, self).step(closure)
 
     def zero_grad(self, set_to_none=False):

@@ -474,13 +474,25 @@ class DLTrainer:
             loss.backward()
             self.backward_time += time.time() - sb
 
-            loss_value = loss.item()
-            self.loss += loss_value
-            self.avg_loss_per_epoch += loss_value
+            if self._async_loss:
+                # device-side accumulation: no per-step .item() host
+                # sync — the loss is read back only at display
+                # boundaries below (SURVEY §2.5.7 notes the reference's
+                # loss.item() drains the pipeline every iteration)
+                if self._loss_accum is None:
+                    self._loss_accum = torch.zeros(
+                        (), dtype=torch.float32, device=self.device)
+                self._loss_accum += loss.detach()
+                loss_value = None
+            else:
+                loss_value = loss.item()
+                self.loss += loss_value
+                self.avg_loss_per_epoch += loss_value
             self.train_iter += 1
         self.times.append(time.time() - s)
 
-        if self.writer is not None and self.rank == 0:
+        if self.writer is not None and self.rank == 0 \
+                and loss_value is not None:
             self.writer.add_scalar('train/loss', loss_value,
                                    self.train_iter)
             self.writer.add_scalar('train/lr', self.lr, self.train_iter)
@@ -494,8 +506,12 @@ class DLTrainer:
             if self._async_loss and self._loss_accum is not None:
                 mean_loss = float(self._loss_accum.item()) / self.display
                 self._loss_accum.zero_()
+                self.avg_loss_per_epoch += mean_loss * self.display
                 logger.info('  mean loss (last %d iters): %.4f',
                             self.display, mean_loss)
+                if self.writer is not None:
+                    self.writer.add_scalar('train/loss', mean_loss,
+                                           self.train_iter)
             logger.info('  phase times (last %d iters): io %.4f fwd %.4f '
                         'bwd %.4f', n, self.io_time / self.display,
                         self.forward_time / self.display,

@@ -39,11 +39,19 @@ class GraphedTrainStep:
         self.static_y = y.clone()
         self.static_loss = None
         # LR through a device buffer: replays follow the LR schedule
-        # without re-capture (set_lr fills it before replay)
+        # without re-capture (set_lr fills it before replay). Without
+        # the fused SGD there is no device LR buffer — a replayed graph
+        # would bake in the capture-time LR (mid warmup-ramp) and
+        # silently train at a frozen tiny LR, so refuse to capture and
+        # let the caller fall back to eager.
         self._fused = getattr(optimizer, '_fused_sgd', None)
-        if self._fused is not None:
-            self._fused.enable_lr_buffer()
-            self._fused.set_lr(optimizer.param_groups[0]['lr'])
+        if self._fused is None:
+            raise RuntimeError(
+                'hipGraph capture needs the fused SGD (device LR '
+                'buffer); the LR schedule would freeze at capture time '
+                'otherwise. Enable MGX_USE_FUSED_SGD or run eager.')
+        self._fused.enable_lr_buffer()
+        self._fused.set_lr(optimizer.param_groups[0]['lr'])
 
         # side-stream warmup, then capture (torch.cuda.graph idiom)
         side = torch.cuda.Stream()

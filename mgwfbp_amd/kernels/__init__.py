@@ -159,6 +159,27 @@ def scale_inplace(buf, scale):
         buf.mul_(scale)
 
 
+_clip_scratch = {}   # device -> 1-elem fp32 norm^2 scratch (stable addr
+                     # so the clip is hipGraph-capturable)
+
+
+def l2norm_clip_(buf, max_norm, eps=1e-6):
+    """Clip ``buf`` to L2 norm <= max_norm entirely on device — no
+    ``.item()`` host round-trip (the reference's clip at
+    distributed_optimizer.py:380-389 host-syncs per merged tensor)."""
+    ext = load_kernels()
+    if ext is not None and buf.is_cuda:
+        scratch = _clip_scratch.get(buf.device)
+        if scratch is None:
+            scratch = torch.zeros(1, dtype=torch.float32,
+                                  device=buf.device)
+            _clip_scratch[buf.device] = scratch
+        ext.l2norm_clip_(buf, scratch, float(max_norm), float(eps))
+    else:
+        coef = torch.clamp(max_norm / (buf.norm(2) + eps), max=1.0)
+        buf.mul_(coef)
+
+
 # --------------------------------------------------------------------------
 # Pure-torch reference implementations (CPU test tier + numerics checks).
 # --------------------------------------------------------------------------

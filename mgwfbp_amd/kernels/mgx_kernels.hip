@@ -309,6 +309,58 @@ __global__ __launch_bounds__(kBlock) void l2norm_sq_kernel(
   }
 }
 
+// Squared-L2 of a flat fp32 buffer (grid-stride, wave64 shuffle reduce,
+// one atomicAdd per block). Feeds clip_scale_kernel: together they form
+// the device-side gradient clip (reference distributed_optimizer.py:
+// 380-389 does norm+clip through host .item() round-trips).
+__global__ __launch_bounds__(kBlock) void l2norm_sq_flat_kernel(
+    const float* __restrict__ buf, long n, float* __restrict__ out) {
+  float acc = 0.f;
+  const long nvec = n & ~3L;
+  const long stride = (long)gridDim.x * blockDim.x * 4L;
+  for (long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4L; i < nvec;
+       i += stride) {
+    float4 v = *reinterpret_cast<const float4*>(buf + i);
+    acc += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+  }
+  const long base = nvec + (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (base < n) acc += buf[base] * buf[base];
+  for (int off = 32; off > 0; off >>= 1)
+    acc += __shfl_down(acc, off, 64);
+  __shared__ float warp_sums[kBlock / 64];
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  if (lane == 0) warp_sums[wid] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float s = 0.f;
+    for (int w = 0; w < kBlock / 64; ++w) s += warp_sums[w];
+    atomicAdd(out, s);
+  }
+}
+
+// buf *= min(1, max_norm / (sqrt(normsq[0]) + eps)) — the clip
+// coefficient is computed per-thread from the device-resident norm, so
+// the whole clip is two launches with zero host round-trips (and is
+// hipGraph-capturable).
+__global__ __launch_bounds__(kBlock) void clip_scale_kernel(
+    float* __restrict__ buf, long n, const float* __restrict__ normsq,
+    float max_norm, float eps) {
+  const float coef =
+      fminf(1.f, max_norm / (sqrtf(normsq[0]) + eps));
+  if (coef >= 1.f) return;
+  const long nvec = n & ~3L;
+  const long stride = (long)gridDim.x * blockDim.x * 4L;
+  for (long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4L; i < nvec;
+       i += stride) {
+    float4 v = *reinterpret_cast<float4*>(buf + i);
+    v.x *= coef; v.y *= coef; v.z *= coef; v.w *= coef;
+    *reinterpret_cast<float4*>(buf + i) = v;
+  }
+  const long base = nvec + (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (base < n) buf[base] *= coef;
+}
+
 __global__ __launch_bounds__(kBlock) void scale_inplace_kernel(
     float* __restrict__ buf, long n, float scale) {
   const long nvec = n & ~3L;
@@ -496,6 +548,27 @@ torch::Tensor l2norm_sq(torch::Tensor t_descs, torch::Tensor t_chunks,
   return out;
 }
 
+void l2norm_clip_(torch::Tensor buf, torch::Tensor normsq_scratch,
+                  double max_norm, double eps) {
+  TORCH_CHECK(buf.is_contiguous() && buf.scalar_type() == at::kFloat);
+  TORCH_CHECK(normsq_scratch.numel() >= 1 &&
+              normsq_scratch.scalar_type() == at::kFloat);
+  const long n = buf.numel();
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const int grid = std::max(
+      1L, std::min((long)kMaxGrid, (n + kBlock * 4 - 1) / (kBlock * 4)));
+  CHECK_HIP(hipMemsetAsync(normsq_scratch.data_ptr(), 0, sizeof(float),
+                           stream));
+  hipLaunchKernelGGL(l2norm_sq_flat_kernel, dim3(grid), dim3(kBlock), 0,
+                     stream, buf.data_ptr<float>(), n,
+                     normsq_scratch.data_ptr<float>());
+  hipLaunchKernelGGL(clip_scale_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                     buf.data_ptr<float>(), n,
+                     normsq_scratch.data_ptr<float>(), (float)max_norm,
+                     (float)eps);
+  CHECK_HIP(hipGetLastError());
+}
+
 void scale_inplace(torch::Tensor buf, double scale) {
   TORCH_CHECK(buf.is_contiguous() && buf.scalar_type() == at::kFloat);
   const long n = buf.numel();
@@ -521,5 +594,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Scatter flat buffer into grads (+cast/scale)");
   m.def("l2norm_sq", &l2norm_sq, "Squared L2 norm over tensor list");
   m.def("scale_inplace", &scale_inplace, "In-place scale of a flat buffer");
+  m.def("l2norm_clip_", &l2norm_clip_,
+        "Device-side L2 norm clip of a flat buffer (no host sync)");
   m.attr("chunk_elems") = py::int_(kChunk);
 }

@@ -184,3 +184,35 @@ class CommunicationProfiler:
             times.append((time.time() - stime) / num_iters)
         logger.debug('comm sweep: %s', list(zip(self.sizes, times)))
         return self.sizes, times
+
+    def benchmark_host_overhead(self, num_calls=100, numel=262144,
+                                dtype=torch.float32):
+        """Mean HOST time to enqueue one async all-reduce.
+
+        Measured launch-to-launch WITHOUT waiting: this is the cost each
+        collective puts on the host thread (python wrapper + RCCL
+        enqueue) even when the device side overlaps perfectly — the
+        per-call constant the MG-WFBP solver amortizes by merging
+        (the serialized sweep above hides it inside
+        one launch+completion round-trip).
+        """
+        device = ('cuda:%d' % torch.cuda.current_device()
+                  if torch.cuda.is_available() else 'cpu')
+        data = torch.randn(numel, dtype=dtype, device=device)
+        # warmup, fully drained
+        for _ in range(5):
+            self.synchronize_fn(self.allreduce_fn(data))
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        handles = []
+        stime = time.time()
+        for _ in range(num_calls):
+            handles.append(self.allreduce_fn(data))
+        host_per_call = (time.time() - stime) / num_calls
+        for h in handles:
+            self.synchronize_fn(h)
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        logger.debug('host enqueue overhead: %.2f us/call',
+                     host_per_call * 1e6)
+        return host_per_call

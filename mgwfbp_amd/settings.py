@@ -48,6 +48,12 @@ FP16 = _env_bool('MGX_FP16', False)              # legacy flag: comm in half pre
 COMM_DTYPE = _env_str('MGX_COMM_DTYPE', 'fp16' if FP16 else 'fp32')
 ADAPTIVE_MERGE = _env_bool('MGX_ADAPTIVE_MERGE', True)   # MG-WFBP solver vs threshold grouping
 ADAPTIVE_ABC = _env_bool('MGX_ADAPTIVE_ABC', True)       # measure alpha/beta online on GPU
+# Per-collective HOST cost (async-enqueue + hook bookkeeping, seconds)
+# added to the solver's per-call constant. On xGMI the device alpha is
+# O(10us) so the host launch path dominates what merging can save; the
+# online sweep measures it (CommunicationProfiler.benchmark_host_overhead)
+# and overrides this default. 0 disables.
+ALPHA_HOST = _env_float('MGX_ALPHA_HOST', 0.0)
 TENSORBOARD = _env_bool('MGX_TENSORBOARD', False)
 MAX_EPOCHS = _env_int('MGX_MAX_EPOCHS', 200)
 USE_HIP_KERNELS = _env_bool('MGX_USE_HIP_KERNELS', True)  # hand-written gfx950 kernels on GPU

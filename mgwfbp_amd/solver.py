@@ -129,12 +129,22 @@ def generate_groups_mgwfbp(
     beta: float,
     nbytes: int = 4,
     size_commtime: 'Tuple[Sequence[float], Sequence[float]]' = None,
+    alpha_host: float = 0.0,
 ) -> Tuple[List[List[str]], Dict[str, int], Dict[str, float]]:
     """Solve the optimal merged-gradient grouping.
 
     Args (forward order): layer names, per-layer backward times (s),
     per-layer element counts; cost-model alpha (s), beta (s/B); bytes per
     element of the comm dtype.
+
+    ``alpha_host`` is the per-collective HOST cost (async-enqueue launch
+    path + hook bookkeeping) that is paid once per group regardless of
+    device-side overlap. On xGMI the device alpha is O(10us) — two
+    orders below the reference's 10GbE table — so amortizing this host
+    launch cost is what merging actually buys; it is added to every
+    collective's cost AND to the startup saving in the merge condition
+    (the reference's ``t_wait < alpha`` test,
+    reference distributed_optimizer.py:239-241).
 
     Returns (groups, key_groupidx_maps, stats). stats carries the solver's
     predicted timeline (reference logs these at
@@ -155,10 +165,16 @@ def generate_groups_mgwfbp(
         t_sizes, t_times = size_commtime
 
         def comm_cost(size_bytes):
-            return predict_from_table(t_sizes, t_times, size_bytes)
+            if size_bytes == 0:
+                return 0.0
+            return alpha_host + predict_from_table(t_sizes, t_times,
+                                                   size_bytes)
     else:
         def comm_cost(size_bytes):
-            return predict_allreduce_time(alpha, beta, size_bytes)
+            if size_bytes == 0:
+                return 0.0
+            return alpha_host + predict_allreduce_time(alpha, beta,
+                                                       size_bytes)
 
     tc = [comm_cost(s * nbytes) for s in p]
     # Gradient-ready offsets: taob[L-1] = 0 (backward starts at the last
@@ -196,7 +212,8 @@ def generate_groups_mgwfbp(
                 merged = True
             else:
                 t_wait = ready_next - taoc[l]
-                if t_wait < alpha:             # waiting < startup saved
+                # waiting < startup saved (device latency + host launch)
+                if t_wait < alpha + alpha_host:
                     merge_into_prev(l)
                     taoc = _comm_start_times(tc, tb, taob, L)
                     merged = True
