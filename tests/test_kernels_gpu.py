@@ -188,3 +188,50 @@ class TestOptimizerGPU:
         for pa, pb in zip(net_a.parameters(), net_b.parameters()):
             assert torch.allclose(pa, pb, atol=1e-5, rtol=1e-5), \
                 (pa - pb).abs().max().item()
+
+
+class TestL2NormClip:
+    """Device-side clip kernels (l2norm_sq_flat + clip_scale) vs the fp32
+    torch reference — VERDICT r01 item 5 (no .item() on the clip path)."""
+
+    @pytest.mark.parametrize('n,scale,max_norm', [
+        (1 << 20, 3.0, 1.5),       # clips
+        (1 << 20, 0.001, 10.0),    # no-op branch
+        (1037, 5.0, 0.25),         # odd tail
+        (4, 2.0, 0.1),             # tiny
+    ])
+    def test_matches_torch_clip(self, kernels, n, scale, max_norm):
+        g = torch.Generator(device='cpu').manual_seed(3)
+        buf = (torch.randn(n, generator=g) * scale).cuda()
+        ref = buf.clone()
+        kernels.l2norm_clip_(buf, max_norm)
+        torch.cuda.synchronize()
+        coef = min(1.0, max_norm / (float(ref.norm(2)) + 1e-6))
+        assert torch.allclose(buf, ref * coef, atol=1e-5, rtol=1e-5)
+        if coef < 1.0:
+            assert float(buf.norm(2)) <= max_norm * (1 + 1e-4)
+
+    def test_no_host_sync_needed(self, kernels):
+        # the wrapper must work mid-stream with pending async work
+        buf = torch.randn(1 << 18, device='cuda') * 10
+        for _ in range(5):
+            buf.add_(0.001)
+            kernels.l2norm_clip_(buf, 2.0)
+        torch.cuda.synchronize()
+        assert float(buf.norm(2)) <= 2.0 * (1 + 1e-4)
+
+
+class TestAsyncLoss:
+    def test_loss_accumulates_on_device(self):
+        """MGX_ASYNC_LOSS: train() must not host-sync per step; the loss
+        lands in a device accumulator read at display boundaries
+        (VERDICT r01 weak #2 — the round-1 flag was dead)."""
+        from mgwfbp_amd.dl_trainer import DLTrainer
+        t = DLTrainer(0, 1, dist=False, batch_size=8, is_weak_scaling=True,
+                      ngpus=1, data_dir='', dataset='cifar10',
+                      dnn='resnet20', lr=0.01, nworkers=1,
+                      prefix='test', synthetic=True)
+        assert t._async_loss
+        t.train(3)
+        assert t._loss_accum is not None
+        assert float(t._loss_accum.item()) > 0.0
