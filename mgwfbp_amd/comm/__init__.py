@@ -182,7 +182,11 @@ class RcclCoreBackend:
                                     timeout=datetime.timedelta(seconds=300))
         self._rank = rk
         self._size = world_size
-        dev = torch.device('cuda', int(os.environ.get('LOCAL_RANK', rk % max(torch.cuda.device_count(), 1))))
+        ndev = max(torch.cuda.device_count(), 1)
+        # modulo so oversubscribed probes (2 ranks x 1 GPU) map onto the
+        # devices that exist
+        dev = torch.device(
+            'cuda', int(os.environ.get('LOCAL_RANK', rk)) % ndev)
         torch.cuda.set_device(dev)
         if world_size > 1:
             if rk == 0:
