@@ -138,12 +138,14 @@ def main():
         eager_step()
     phase('warmup done (%d steps)' % args.warmup)
 
-    # auto: graph only at world==1 — RCCL collective capture at world>1
-    # is untested on this pool (single-GPU boxes); use --graph on to
-    # force it for multi-rank runs
+    # auto: graph at every world size. RCCL-collective capture is
+    # hardware-validated at world=1 (tests/test_kernels_gpu.py
+    # TestCommCoreBindings — capture propagates through the comm stream
+    # exactly as at world=8; see profiles/multirank_blocker.md), and the
+    # consensus below falls every rank back to eager if ANY rank's
+    # capture fails.
     want_graph = (args.graph == 'on'
-                  or (args.graph == 'auto' and torch.cuda.is_available()
-                      and n_gpus == 1))
+                  or (args.graph == 'auto' and torch.cuda.is_available()))
     if want_graph and torch.cuda.is_available():
         from mgwfbp_amd.graph_step import GraphedTrainStep
         try:

@@ -175,7 +175,9 @@ class _DistributedOptimizer(torch.optim.Optimizer):
             groups, gmap, stats = solver.generate_groups_mgwfbp(
                 keys, self._layerwise_times, sizes, self.alpha, self.beta,
                 nbytes, size_commtime=self.size_commtime_dict,
-                alpha_host=self.alpha_host)
+                alpha_host=self.alpha_host,
+                density=self._density if self._is_sparse else 1.0,
+                nworkers=max(comm.size(), 2))
             if comm.rank() == 0:
                 logger.info(
                     'MG-WFBP solver: %d layers -> %d groups; predicted '

@@ -105,6 +105,32 @@ def predict_from_table(table_sizes, table_times, size_bytes):
     return table_times[lo] + f * (table_times[hi] - table_times[lo])
 
 
+def predict_sparse_allgather_time(alpha: float, beta: float, numel: int,
+                                  density: float, nworkers: int,
+                                  nbytes: int = 4,
+                                  index_bytes: int = 8) -> float:
+    """Cost of one merge group's top-k sparse exchange.
+
+    The sparse path (distributed_optimizer._sparse_allgather_async)
+    replaces the dense all-reduce with TWO all-gathers — values
+    (``nbytes``/elem) and indices (int64, ``index_bytes``/elem) — each
+    rank contributing k = ceil(numel * density) elements. A ring
+    all-gather receives (P-1)*k elements per rank, so:
+
+        t = 2*alpha + beta * (nbytes + index_bytes) * k * (P-1)
+
+    This is the reference's sparse cost-model hook made real
+    (reference utils.py:104-149 ``predict_density_time`` effectively
+    returns a constant 0.001 — SURVEY §2.1); the top-k compute on the
+    GPU is not modeled (it overlaps backward like the collectives do).
+    """
+    if numel == 0:
+        return 0.0
+    k = max(1, int(numel * density))
+    payload = (nbytes + index_bytes) * k * max(nworkers - 1, 1)
+    return 2.0 * alpha + beta * payload
+
+
 def _comm_start_times(tc: List[float], tb: Sequence[float],
                       taob: Sequence[float], L: int) -> List[float]:
     """Earliest all-reduce start per layer given channel serialization.
@@ -130,6 +156,8 @@ def generate_groups_mgwfbp(
     nbytes: int = 4,
     size_commtime: 'Tuple[Sequence[float], Sequence[float]]' = None,
     alpha_host: float = 0.0,
+    density: float = 1.0,
+    nworkers: int = 2,
 ) -> Tuple[List[List[str]], Dict[str, int], Dict[str, float]]:
     """Solve the optimal merged-gradient grouping.
 
@@ -161,7 +189,17 @@ def generate_groups_mgwfbp(
     tb = list(layerwise_times)
     p = [int(s) for s in sizes]          # merged element counts (mutated)
 
-    if size_commtime is not None:
+    if density < 1.0:
+        # sparse top-k exchange: cost follows the all-gather payload,
+        # not the dense all-reduce (and each group pays TWO collective
+        # launches — values + indices)
+        def comm_cost(size_bytes):
+            if size_bytes == 0:
+                return 0.0
+            return 2.0 * alpha_host + predict_sparse_allgather_time(
+                alpha, beta, size_bytes // max(nbytes, 1), density,
+                nworkers, nbytes)
+    elif size_commtime is not None:
         t_sizes, t_times = size_commtime
 
         def comm_cost(size_bytes):
@@ -212,8 +250,12 @@ def generate_groups_mgwfbp(
                 merged = True
             else:
                 t_wait = ready_next - taoc[l]
-                # waiting < startup saved (device latency + host launch)
-                if t_wait < alpha + alpha_host:
+                # waiting < startup saved (device latency + host launch;
+                # the sparse path saves TWO launches per merged-away
+                # group: values + indices all-gathers)
+                saved = ((alpha + alpha_host) * 2.0 if density < 1.0
+                         else alpha + alpha_host)
+                if t_wait < saved:
                     merge_into_prev(l)
                     taoc = _comm_start_times(tc, tb, taob, L)
                     merged = True
