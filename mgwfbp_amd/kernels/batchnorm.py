@@ -50,6 +50,31 @@ class _BatchNormFunc(torch.autograd.Function):
         return dx, dgamma, dbeta, None, None, None, None, None
 
 
+class _BatchNormAddReLUFunc(torch.autograd.Function):
+    """y = relu(bn(x) + residual) in one normalize pass; backward gates
+    dy by the recomputed post-add sign and emits the residual gradient
+    in the same dx kernel (bn_kernels.hip ADD path)."""
+
+    @staticmethod
+    def forward(ctx, x, residual, weight, bias, running_mean, running_var,
+                momentum, eps):
+        ext = _load()
+        y, mean, invstd = ext.bn_fwd_train(
+            x, weight, bias, running_mean, running_var, momentum, eps,
+            True, residual)
+        ctx.save_for_backward(x, residual, weight, bias, mean, invstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _load()
+        x, residual, weight, bias, mean, invstd = ctx.saved_tensors
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        dx, dgamma, dbeta, dres = ext.bn_bwd(dy, x, mean, invstd, weight,
+                                             bias, True, residual)
+        return (dx, dres, dgamma, dbeta, None, None, None, None)
+
+
 class MgxBatchNorm2d(nn.BatchNorm2d):
     fuse_relu = False
 
@@ -80,6 +105,34 @@ class MgxBatchNorm2d(nn.BatchNorm2d):
                                self.running_mean, self.running_var,
                                self.eps, self.fuse_relu)
 
+    def forward_add_relu(self, x, residual):
+        """Fused y = relu(bn(x) + residual) (the ResNet post-add
+        activation). Falls back to the unfused chain when the HIP path
+        can't run (CPU, odd layout)."""
+        if not (self._use_hip_path(x) and residual.is_cuda
+                and residual.shape == x.shape):
+            y = nn.BatchNorm2d.forward(self, x)
+            return torch.nn.functional.relu(y + residual, inplace=True)
+        if residual.dtype != x.dtype:
+            # autocast leaves the downsample BN's output in fp32 while
+            # the conv output is bf16: one cast beats losing the fusion
+            residual = residual.to(x.dtype)
+        residual = residual.contiguous(
+            memory_format=torch.channels_last)
+        if self.training:
+            if self.num_batches_tracked is not None:
+                self.num_batches_tracked.add_(1)
+            momentum = self.momentum
+            if momentum is None:
+                momentum = 1.0 / float(self.num_batches_tracked)
+            return _BatchNormAddReLUFunc.apply(
+                x, residual, self.weight, self.bias, self.running_mean,
+                self.running_var, momentum, self.eps)
+        ext = _load()
+        return ext.bn_fwd_eval(x, self.weight, self.bias,
+                               self.running_mean, self.running_var,
+                               self.eps, True, residual)
+
 
 def _mgx_from(child, fuse_relu=False):
     bn = MgxBatchNorm2d(child.num_features, eps=child.eps,
@@ -108,11 +161,11 @@ def convert_batchnorm(module, fuse_relu=True, only_fused=False):
     more than the relu fusion saves on big models (measured A/B,
     profiles/README.md).
     """
-    from ..models.common import BNReLU
+    from ..models.common import BNReLU, BNAddReLU
     children = list(module.named_children())
     names = [n for n, _ in children]
     for idx, (name, child) in enumerate(children):
-        if isinstance(child, BNReLU):
+        if isinstance(child, (BNReLU, BNAddReLU)):
             if type(child.bn) is nn.BatchNorm2d and fuse_relu:
                 child.bn = _mgx_from(child.bn, fuse_relu=True)
         elif type(child) is nn.BatchNorm2d:

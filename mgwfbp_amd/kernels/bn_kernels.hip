@@ -287,12 +287,16 @@ __global__ __launch_bounds__(kBlock) void bn_fwd_finalize_kernel(
 }
 
 // --------------------------------------------------- fwd: norm ----------
-template <typename T>
+// ADD: y = [relu](bn(x) + res) — the ResNet post-add activation folded
+// into the normalize pass: ONE kernel replaces bn_out store + add read/
+// read/write + relu read/write (4 extra full-tensor HBM passes in the
+// unfused eager chain). res is loaded once per element alongside x.
+template <typename T, bool ADD>
 __global__ __launch_bounds__(kBlock) void bn_fwd_norm_kernel(
-    const T* __restrict__ x, T* __restrict__ y, long rows, long C,
-    const float* __restrict__ mean, const float* __restrict__ invstd,
-    const float* __restrict__ gamma, const float* __restrict__ beta,
-    bool relu) {
+    const T* __restrict__ x, const T* __restrict__ res, T* __restrict__ y,
+    long rows, long C, const float* __restrict__ mean,
+    const float* __restrict__ invstd, const float* __restrict__ gamma,
+    const float* __restrict__ beta, bool relu) {
   int opr, rpb, o, rl; bool active;
   tile_map(C, opr, rpb, o, rl, active);
   if (!active) return;
@@ -309,13 +313,21 @@ __global__ __launch_bounds__(kBlock) void bn_fwd_norm_kernel(
     const long rstride = (long)gridDim.x * rpb;
     long r = (long)blockIdx.x * rpb + rl;
     for (; r + rstride < rows; r += 2 * rstride) {
-      float v[8], w[8];
+      float v[8], w[8], a[8], b[8];
       VecIO<T>::load(x + r * C + c0, v);
       VecIO<T>::load(x + (r + rstride) * C + c0, w);
+      if (ADD) {
+        VecIO<T>::load(res + r * C + c0, a);
+        VecIO<T>::load(res + (r + rstride) * C + c0, b);
+      }
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         v[i] = v[i] * sc[i] + sh[i];
         w[i] = w[i] * sc[i] + sh[i];
+        if (ADD) {
+          v[i] += a[i];
+          w[i] += b[i];
+        }
         if (relu) {
           v[i] = v[i] > 0.f ? v[i] : 0.f;
           w[i] = w[i] > 0.f ? w[i] : 0.f;
@@ -325,11 +337,13 @@ __global__ __launch_bounds__(kBlock) void bn_fwd_norm_kernel(
       VecIO<T>::store(y + (r + rstride) * C + c0, w);
     }
     for (; r < rows; r += rstride) {
-      float v[8];
+      float v[8], a[8];
       VecIO<T>::load(x + r * C + c0, v);
+      if (ADD) VecIO<T>::load(res + r * C + c0, a);
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         v[i] = v[i] * sc[i] + sh[i];
+        if (ADD) v[i] += a[i];
         if (relu) v[i] = v[i] > 0.f ? v[i] : 0.f;
       }
       VecIO<T>::store(y + r * C + c0, v);
@@ -340,10 +354,12 @@ __global__ __launch_bounds__(kBlock) void bn_fwd_norm_kernel(
 // -------------------------------------------------- bwd: reduce ---------
 // RELU: the forward fused y = relu(bn(x)); backward gates dy by y>0,
 // recomputing y's sign from (x, mean, invstd, gamma, beta) — no saved
-// activation needed.
-template <typename T, bool RELU>
+// activation needed. ADD: forward was y = relu(bn(x) + res); the gate
+// recomputes bn(x)+res's sign, loading res alongside x.
+template <typename T, bool RELU, bool ADD>
 __global__ __launch_bounds__(kBlock) void bn_bwd_reduce_kernel(
-    const T* __restrict__ dy, const T* __restrict__ x, long rows, long C,
+    const T* __restrict__ dy, const T* __restrict__ x,
+    const T* __restrict__ res, long rows, long C,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     float* __restrict__ partial) {
@@ -371,31 +387,40 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_reduce_kernel(
       const long rstride = (long)gridDim.x * rpb;
       long r = (long)blockIdx.x * rpb + rl;
       for (; r + rstride < rows; r += 2 * rstride) {  // 4 loads in flight
-        float g0[8], v0[8], g1[8], v1[8];
+        float g0[8], v0[8], g1[8], v1[8], a0[8], a1[8];
         VecIO<T>::load(dy + r * C + c0, g0);
         VecIO<T>::load(x + r * C + c0, v0);
         VecIO<T>::load(dy + (r + rstride) * C + c0, g1);
         VecIO<T>::load(x + (r + rstride) * C + c0, v1);
+        if (ADD) {
+          VecIO<T>::load(res + r * C + c0, a0);
+          VecIO<T>::load(res + (r + rstride) * C + c0, a1);
+        }
 #pragma unroll
         for (int i = 0; i < 8; ++i) {
           const float xh0 = (v0[i] - mu[i]) * is[i];
           const float xh1 = (v1[i] - mu[i]) * is[i];
           if (RELU) {
-            if (xh0 * ga[i] + be[i] <= 0.f) g0[i] = 0.f;
-            if (xh1 * ga[i] + be[i] <= 0.f) g1[i] = 0.f;
+            const float r0 = ADD ? a0[i] : 0.f;
+            const float r1 = ADD ? a1[i] : 0.f;
+            if (xh0 * ga[i] + be[i] + r0 <= 0.f) g0[i] = 0.f;
+            if (xh1 * ga[i] + be[i] + r1 <= 0.f) g1[i] = 0.f;
           }
           sd[i] += g0[i] + g1[i];
           sx[i] += g0[i] * xh0 + g1[i] * xh1;
         }
       }
       for (; r < rows; r += rstride) {
-        float g[8], v[8];
+        float g[8], v[8], a[8];
         VecIO<T>::load(dy + r * C + c0, g);
         VecIO<T>::load(x + r * C + c0, v);
+        if (ADD) VecIO<T>::load(res + r * C + c0, a);
 #pragma unroll
         for (int i = 0; i < 8; ++i) {
           const float xh = (v[i] - mu[i]) * is[i];
-          if (RELU && xh * ga[i] + be[i] <= 0.f) g[i] = 0.f;
+          if (RELU &&
+              xh * ga[i] + be[i] + (ADD ? a[i] : 0.f) <= 0.f)
+            g[i] = 0.f;
           sd[i] += g[i];
           sx[i] += g[i] * xh;
         }
@@ -429,9 +454,13 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_finalize_kernel(
 
 // ---------------------------------------------------- bwd: dx -----------
 // dx = gamma*invstd*(dy - dbeta/M - xhat*dgamma/M)
-template <typename T, bool RELU>
+// ADD: dy is first gated by the recomputed post-add sign; the gated dy
+// IS the residual branch's gradient, stored to dres in the same pass
+// (saving the separate relu-backward kernel of the unfused chain).
+template <typename T, bool RELU, bool ADD>
 __global__ __launch_bounds__(kBlock) void bn_bwd_dx_kernel(
-    const T* __restrict__ dy, const T* __restrict__ x, T* __restrict__ dx,
+    const T* __restrict__ dy, const T* __restrict__ x,
+    const T* __restrict__ res, T* __restrict__ dx, T* __restrict__ dres,
     long rows, long C, const float* __restrict__ mean,
     const float* __restrict__ invstd, const float* __restrict__ gamma,
     const float* __restrict__ beta, const float* __restrict__ dbeta,
@@ -456,35 +485,53 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_dx_kernel(
     const long rstride = (long)gridDim.x * rpb;
     long r = (long)blockIdx.x * rpb + rl;
     for (; r + rstride < rows; r += 2 * rstride) {
-      float g0[8], v0[8], g1[8], v1[8];
+      float g0[8], v0[8], g1[8], v1[8], a0[8], a1[8];
       VecIO<T>::load(dy + r * C + c0, g0);
       VecIO<T>::load(x + r * C + c0, v0);
       VecIO<T>::load(dy + (r + rstride) * C + c0, g1);
       VecIO<T>::load(x + (r + rstride) * C + c0, v1);
+      if (ADD) {
+        VecIO<T>::load(res + r * C + c0, a0);
+        VecIO<T>::load(res + (r + rstride) * C + c0, a1);
+      }
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         const float xh0 = (v0[i] - mu[i]) * is[i];
         const float xh1 = (v1[i] - mu[i]) * is[i];
         if (RELU) {
-          if (xh0 * ga[i] + be[i] <= 0.f) g0[i] = 0.f;
-          if (xh1 * ga[i] + be[i] <= 0.f) g1[i] = 0.f;
+          const float r0 = ADD ? a0[i] : 0.f;
+          const float r1 = ADD ? a1[i] : 0.f;
+          if (xh0 * ga[i] + be[i] + r0 <= 0.f) g0[i] = 0.f;
+          if (xh1 * ga[i] + be[i] + r1 <= 0.f) g1[i] = 0.f;
+        }
+        if (ADD) {
+          a0[i] = g0[i];   // gated dy == residual gradient
+          a1[i] = g1[i];
         }
         g0[i] = gi[i] * (g0[i] - md[i] - xh0 * mx[i]);
         g1[i] = gi[i] * (g1[i] - md[i] - xh1 * mx[i]);
+      }
+      if (ADD) {
+        VecIO<T>::store(dres + r * C + c0, a0);
+        VecIO<T>::store(dres + (r + rstride) * C + c0, a1);
       }
       VecIO<T>::store(dx + r * C + c0, g0);
       VecIO<T>::store(dx + (r + rstride) * C + c0, g1);
     }
     for (; r < rows; r += rstride) {
-      float g[8], v[8];
+      float g[8], v[8], a[8];
       VecIO<T>::load(dy + r * C + c0, g);
       VecIO<T>::load(x + r * C + c0, v);
+      if (ADD) VecIO<T>::load(res + r * C + c0, a);
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         const float xh = (v[i] - mu[i]) * is[i];
-        if (RELU && xh * ga[i] + be[i] <= 0.f) g[i] = 0.f;
+        if (RELU && xh * ga[i] + be[i] + (ADD ? a[i] : 0.f) <= 0.f)
+          g[i] = 0.f;
+        if (ADD) a[i] = g[i];
         g[i] = gi[i] * (g[i] - md[i] - xh * mx[i]);
       }
+      if (ADD) VecIO<T>::store(dres + r * C + c0, a);
       VecIO<T>::store(dx + r * C + c0, g);
     }
   }
@@ -561,7 +608,9 @@ std::pair<float*, long> finalize_tree(torch::Tensor& partial, int rblocks,
 std::vector<torch::Tensor> bn_fwd_train(
     torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
     torch::Tensor running_mean, torch::Tensor running_var, double momentum,
-    double eps, bool relu) {
+    double eps, bool relu, c10::optional<torch::Tensor> residual_opt) {
+  torch::Tensor residual =
+      residual_opt.has_value() ? residual_opt.value() : torch::Tensor();
   auto g = geom(x);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   auto opts = torch::TensorOptions().dtype(torch::kFloat).device(x.device());
@@ -572,6 +621,13 @@ std::vector<torch::Tensor> bn_fwd_train(
   auto mean = torch::empty({g.C}, opts);
   auto invstd = torch::empty({g.C}, opts);
   auto y = torch::empty_like(x);
+  const bool add = residual.defined();
+  if (add) {
+    TORCH_CHECK(residual.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                    residual.sizes() == x.sizes() &&
+                    residual.scalar_type() == x.scalar_type(),
+                "residual must match x (channels_last, shape, dtype)");
+  }
   DISPATCH_DT(x.scalar_type(), {
     hipLaunchKernelGGL(bn_fwd_reduce_kernel<dt>, dim3(rblocks),
                        dim3(kBlock), 0, stream,
@@ -589,14 +645,28 @@ std::vector<torch::Tensor> bn_fwd_train(
                      running_var.defined()
                          ? running_var.data_ptr<float>() : nullptr);
   DISPATCH_DT(x.scalar_type(), {
-    hipLaunchKernelGGL(bn_fwd_norm_kernel<dt>, dim3(eblocks), dim3(kBlock),
-                       0, stream,
-                       reinterpret_cast<const dt*>(x.data_ptr()),
-                       reinterpret_cast<dt*>(y.data_ptr()), g.rows, g.C,
-                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       gamma.defined() ? gamma.data_ptr<float>() : nullptr,
-                       beta.defined() ? beta.data_ptr<float>() : nullptr,
-                       relu);
+    if (add)
+      hipLaunchKernelGGL((bn_fwd_norm_kernel<dt, true>), dim3(eblocks),
+                         dim3(kBlock), 0, stream,
+                         reinterpret_cast<const dt*>(x.data_ptr()),
+                         reinterpret_cast<const dt*>(residual.data_ptr()),
+                         reinterpret_cast<dt*>(y.data_ptr()), g.rows, g.C,
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         gamma.defined() ? gamma.data_ptr<float>()
+                                         : nullptr,
+                         beta.defined() ? beta.data_ptr<float>() : nullptr,
+                         relu);
+    else
+      hipLaunchKernelGGL((bn_fwd_norm_kernel<dt, false>), dim3(eblocks),
+                         dim3(kBlock), 0, stream,
+                         reinterpret_cast<const dt*>(x.data_ptr()),
+                         nullptr,
+                         reinterpret_cast<dt*>(y.data_ptr()), g.rows, g.C,
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         gamma.defined() ? gamma.data_ptr<float>()
+                                         : nullptr,
+                         beta.defined() ? beta.data_ptr<float>() : nullptr,
+                         relu);
   });
   CHECK_HIP(hipGetLastError());
   return {y, mean, invstd};
@@ -605,22 +675,41 @@ std::vector<torch::Tensor> bn_fwd_train(
 torch::Tensor bn_fwd_eval(torch::Tensor x, torch::Tensor gamma,
                           torch::Tensor beta, torch::Tensor running_mean,
                           torch::Tensor running_var, double eps,
-                          bool relu) {
+                          bool relu,
+                          c10::optional<torch::Tensor> residual_opt) {
+  torch::Tensor residual =
+      residual_opt.has_value() ? residual_opt.value() : torch::Tensor();
   auto g = geom(x);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   auto invstd = (running_var + eps).rsqrt();
   auto y = torch::empty_like(x);
   const int eblocks = grid_elem(g);
+  const bool add = residual.defined();
   DISPATCH_DT(x.scalar_type(), {
-    hipLaunchKernelGGL(bn_fwd_norm_kernel<dt>, dim3(eblocks), dim3(kBlock),
-                       0, stream,
-                       reinterpret_cast<const dt*>(x.data_ptr()),
-                       reinterpret_cast<dt*>(y.data_ptr()), g.rows, g.C,
-                       running_mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(),
-                       gamma.defined() ? gamma.data_ptr<float>() : nullptr,
-                       beta.defined() ? beta.data_ptr<float>() : nullptr,
-                       relu);
+    if (add)
+      hipLaunchKernelGGL((bn_fwd_norm_kernel<dt, true>), dim3(eblocks),
+                         dim3(kBlock), 0, stream,
+                         reinterpret_cast<const dt*>(x.data_ptr()),
+                         reinterpret_cast<const dt*>(residual.data_ptr()),
+                         reinterpret_cast<dt*>(y.data_ptr()), g.rows, g.C,
+                         running_mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(),
+                         gamma.defined() ? gamma.data_ptr<float>()
+                                         : nullptr,
+                         beta.defined() ? beta.data_ptr<float>() : nullptr,
+                         relu);
+    else
+      hipLaunchKernelGGL((bn_fwd_norm_kernel<dt, false>), dim3(eblocks),
+                         dim3(kBlock), 0, stream,
+                         reinterpret_cast<const dt*>(x.data_ptr()),
+                         nullptr,
+                         reinterpret_cast<dt*>(y.data_ptr()), g.rows, g.C,
+                         running_mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(),
+                         gamma.defined() ? gamma.data_ptr<float>()
+                                         : nullptr,
+                         beta.defined() ? beta.data_ptr<float>() : nullptr,
+                         relu);
   });
   CHECK_HIP(hipGetLastError());
   return y;
@@ -629,7 +718,10 @@ torch::Tensor bn_fwd_eval(torch::Tensor x, torch::Tensor gamma,
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor mean, torch::Tensor invstd,
                                   torch::Tensor gamma, torch::Tensor beta,
-                                  bool relu) {
+                                  bool relu,
+                                  c10::optional<torch::Tensor> residual_opt) {
+  torch::Tensor residual =
+      residual_opt.has_value() ? residual_opt.value() : torch::Tensor();
   auto g = geom(x);
   TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast),
               "expected channels_last grad");
@@ -642,23 +734,37 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   auto dbeta = torch::empty({g.C}, opts);
   auto dgamma = torch::empty({g.C}, opts);
   auto dx = torch::empty_like(x);
+  const bool add = residual.defined();
+  TORCH_CHECK(!add || relu, "residual fusion implies post-add relu");
+  torch::Tensor dres;
   const float* gamma_p = gamma.defined() ? gamma.data_ptr<float>()
                                          : nullptr;
   const float* beta_p = beta.defined() ? beta.data_ptr<float>() : nullptr;
   DISPATCH_DT(x.scalar_type(), {
-    if (relu)
-      hipLaunchKernelGGL((bn_bwd_reduce_kernel<dt, true>), dim3(rblocks),
-                         dim3(kBlock), 0, stream,
+    if (add)
+      hipLaunchKernelGGL((bn_bwd_reduce_kernel<dt, true, true>),
+                         dim3(rblocks), dim3(kBlock), 0, stream,
                          reinterpret_cast<const dt*>(dy.data_ptr()),
-                         reinterpret_cast<const dt*>(x.data_ptr()), g.rows,
+                         reinterpret_cast<const dt*>(x.data_ptr()),
+                         reinterpret_cast<const dt*>(residual.data_ptr()),
+                         g.rows, g.C, mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(), gamma_p, beta_p,
+                         partial.data_ptr<float>());
+    else if (relu)
+      hipLaunchKernelGGL((bn_bwd_reduce_kernel<dt, true, false>),
+                         dim3(rblocks), dim3(kBlock), 0, stream,
+                         reinterpret_cast<const dt*>(dy.data_ptr()),
+                         reinterpret_cast<const dt*>(x.data_ptr()),
+                         nullptr, g.rows,
                          g.C, mean.data_ptr<float>(),
                          invstd.data_ptr<float>(), gamma_p, beta_p,
                          partial.data_ptr<float>());
     else
-      hipLaunchKernelGGL((bn_bwd_reduce_kernel<dt, false>), dim3(rblocks),
-                         dim3(kBlock), 0, stream,
+      hipLaunchKernelGGL((bn_bwd_reduce_kernel<dt, false, false>),
+                         dim3(rblocks), dim3(kBlock), 0, stream,
                          reinterpret_cast<const dt*>(dy.data_ptr()),
-                         reinterpret_cast<const dt*>(x.data_ptr()), g.rows,
+                         reinterpret_cast<const dt*>(x.data_ptr()),
+                         nullptr, g.rows,
                          g.C, mean.data_ptr<float>(),
                          invstd.data_ptr<float>(), gamma_p, beta_p,
                          partial.data_ptr<float>());
@@ -669,26 +775,44 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                      stream, finb.first, finb.second, g.C,
                      dbeta.data_ptr<float>(), dgamma.data_ptr<float>());
   DISPATCH_DT(x.scalar_type(), {
-    if (relu)
-      hipLaunchKernelGGL((bn_bwd_dx_kernel<dt, true>), dim3(eblocks),
-                         dim3(kBlock), 0, stream,
+    if (add) {
+      dres = torch::empty_like(x);
+      hipLaunchKernelGGL((bn_bwd_dx_kernel<dt, true, true>),
+                         dim3(eblocks), dim3(kBlock), 0, stream,
                          reinterpret_cast<const dt*>(dy.data_ptr()),
                          reinterpret_cast<const dt*>(x.data_ptr()),
-                         reinterpret_cast<dt*>(dx.data_ptr()), g.rows, g.C,
+                         reinterpret_cast<const dt*>(residual.data_ptr()),
+                         reinterpret_cast<dt*>(dx.data_ptr()),
+                         reinterpret_cast<dt*>(dres.data_ptr()),
+                         g.rows, g.C,
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         gamma_p, beta_p, dbeta.data_ptr<float>(),
+                         dgamma.data_ptr<float>(), 1.f / (float)g.rows);
+    } else if (relu)
+      hipLaunchKernelGGL((bn_bwd_dx_kernel<dt, true, false>),
+                         dim3(eblocks), dim3(kBlock), 0, stream,
+                         reinterpret_cast<const dt*>(dy.data_ptr()),
+                         reinterpret_cast<const dt*>(x.data_ptr()),
+                         nullptr,
+                         reinterpret_cast<dt*>(dx.data_ptr()), nullptr,
+                         g.rows, g.C,
                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
                          gamma_p, beta_p, dbeta.data_ptr<float>(),
                          dgamma.data_ptr<float>(), 1.f / (float)g.rows);
     else
-      hipLaunchKernelGGL((bn_bwd_dx_kernel<dt, false>), dim3(eblocks),
-                         dim3(kBlock), 0, stream,
+      hipLaunchKernelGGL((bn_bwd_dx_kernel<dt, false, false>),
+                         dim3(eblocks), dim3(kBlock), 0, stream,
                          reinterpret_cast<const dt*>(dy.data_ptr()),
                          reinterpret_cast<const dt*>(x.data_ptr()),
-                         reinterpret_cast<dt*>(dx.data_ptr()), g.rows, g.C,
+                         nullptr,
+                         reinterpret_cast<dt*>(dx.data_ptr()), nullptr,
+                         g.rows, g.C,
                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
                          gamma_p, beta_p, dbeta.data_ptr<float>(),
                          dgamma.data_ptr<float>(), 1.f / (float)g.rows);
   });
   CHECK_HIP(hipGetLastError());
+  if (add) return {dx, dgamma, dbeta, dres};
   return {dx, dgamma, dbeta};
 }
 
@@ -696,8 +820,19 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd_train", &bn_fwd_train,
-        "NHWC BN training forward; returns (y, mean, invstd)");
-  m.def("bn_fwd_eval", &bn_fwd_eval, "NHWC BN inference forward");
+        "NHWC BN training forward (optionally y=relu(bn(x)+residual)); "
+        "returns (y, mean, invstd)",
+        py::arg("x"), py::arg("gamma"), py::arg("beta"),
+        py::arg("running_mean"), py::arg("running_var"),
+        py::arg("momentum"), py::arg("eps"), py::arg("relu"),
+        py::arg("residual") = py::none());
+  m.def("bn_fwd_eval", &bn_fwd_eval, "NHWC BN inference forward",
+        py::arg("x"), py::arg("gamma"), py::arg("beta"),
+        py::arg("running_mean"), py::arg("running_var"), py::arg("eps"),
+        py::arg("relu"), py::arg("residual") = py::none());
   m.def("bn_bwd", &bn_bwd,
-        "NHWC BN backward; returns (dx, dgamma, dbeta)");
+        "NHWC BN backward; returns (dx, dgamma, dbeta[, dresidual])",
+        py::arg("dy"), py::arg("x"), py::arg("mean"), py::arg("invstd"),
+        py::arg("gamma"), py::arg("beta"), py::arg("relu"),
+        py::arg("residual") = py::none());
 }

@@ -10,7 +10,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from .common import BNReLU
+from .common import BNReLU, BNAddReLU
 
 
 def _conv3x3(cin, cout, stride=1):
@@ -39,7 +39,7 @@ class BasicBlock(nn.Module):
         self.conv1 = _conv3x3(cin, cout, stride)
         self.bn1 = BNReLU(cout)
         self.conv2 = _conv3x3(cout, cout)
-        self.bn2 = nn.BatchNorm2d(cout)   # relu after the add
+        self.bn2 = BNAddReLU(cout)   # bn + residual add + relu, fused
         self.shortcut = nn.Sequential()
         if stride != 1 or cin != cout:
             if option == 'A':
@@ -52,9 +52,7 @@ class BasicBlock(nn.Module):
 
     def forward(self, x):
         out = self.bn1(self.conv1(x))
-        out = self.bn2(self.conv2(out))
-        out = out + self.shortcut(x)
-        return F.relu(out)
+        return self.bn2(self.conv2(out), self.shortcut(x))
 
 
 class CifarResNet(nn.Module):

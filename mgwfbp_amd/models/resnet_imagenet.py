@@ -7,7 +7,7 @@ torchvision is not part of this framework's dependency set.
 import torch.nn as nn
 import torch.nn.functional as F
 
-from .common import BNReLU
+from .common import BNReLU, BNAddReLU
 
 
 class BasicBlock(nn.Module):
@@ -18,14 +18,13 @@ class BasicBlock(nn.Module):
         self.conv1 = nn.Conv2d(cin, width, 3, stride, 1, bias=False)
         self.bn1 = BNReLU(width)
         self.conv2 = nn.Conv2d(width, width, 3, 1, 1, bias=False)
-        self.bn2 = nn.BatchNorm2d(width)   # relu comes after the add
+        self.bn2 = BNAddReLU(width)   # bn + residual add + relu, fused
         self.downsample = downsample
 
     def forward(self, x):
         identity = x if self.downsample is None else self.downsample(x)
         out = self.bn1(self.conv1(x))
-        out = self.bn2(self.conv2(out))
-        return F.relu(out + identity, inplace=True)
+        return self.bn2(self.conv2(out), identity)
 
 
 class Bottleneck(nn.Module):
@@ -38,15 +37,14 @@ class Bottleneck(nn.Module):
         self.conv2 = nn.Conv2d(width, width, 3, stride, 1, bias=False)
         self.bn2 = BNReLU(width)
         self.conv3 = nn.Conv2d(width, width * 4, 1, bias=False)
-        self.bn3 = nn.BatchNorm2d(width * 4)   # relu after the add
+        self.bn3 = BNAddReLU(width * 4)   # bn + residual add + relu
         self.downsample = downsample
 
     def forward(self, x):
         identity = x if self.downsample is None else self.downsample(x)
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
-        out = self.bn3(self.conv3(out))
-        return F.relu(out + identity, inplace=True)
+        return self.bn3(self.conv3(out), identity)
 
 
 class ResNet(nn.Module):

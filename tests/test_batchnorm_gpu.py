@@ -142,3 +142,96 @@ def test_convert_fuses_bnrelu_units_and_sequentials():
     fused_v = sum(1 for m in vgg.modules()
                   if isinstance(m, MgxBatchNorm2d) and m.fuse_relu)
     assert fused_v == 13, fused_v
+
+
+@pytest.mark.parametrize('C', [64, 256, 2048])
+@pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16])
+def test_fused_bn_add_relu_matches_composite(C, dtype):
+    """y = relu(bn(x) + residual) fused (BNAddReLU epilogue, VERDICT r01
+    item 7) vs the fp32 torch chain, incl. the residual gradient the dx
+    kernel emits in-pass."""
+    from mgwfbp_amd.kernels.batchnorm import MgxBatchNorm2d
+    torch.manual_seed(3)
+    ref = nn.BatchNorm2d(C).cuda()
+    ref.weight.data.uniform_(0.5, 1.5)
+    ref.bias.data.uniform_(-0.5, 0.5)
+    ours = MgxBatchNorm2d(C).cuda()
+    ours.load_state_dict(ref.state_dict())
+    ours.fuse_relu = True
+
+    x1 = _mk(C, dtype=dtype).requires_grad_(True)
+    r1 = _mk(C, dtype=dtype, seed=9).requires_grad_(True)
+    x2 = x1.detach().float().clone().requires_grad_(True)
+    r2 = r1.detach().float().clone().requires_grad_(True)
+
+    y_ref = torch.nn.functional.relu(ref(x2) + r2)
+    y = ours.forward_add_relu(x1, r1)
+    tol = 2e-5 if dtype == torch.float32 else 5e-2
+    assert torch.allclose(y.float(), y_ref, atol=tol, rtol=tol), \
+        (y.float() - y_ref).abs().max().item()
+
+    dy = torch.randn_like(y_ref)
+    y_ref.backward(dy)
+    y.backward(dy.to(dtype).to(memory_format=torch.channels_last))
+    assert torch.allclose(x1.grad.float(), x2.grad, atol=tol * 5,
+                          rtol=tol * 5), \
+        (x1.grad.float() - x2.grad).abs().max().item()
+    assert torch.allclose(r1.grad.float(), r2.grad, atol=tol * 5,
+                          rtol=tol * 5), \
+        (r1.grad.float() - r2.grad).abs().max().item()
+    gatol, grtol = (1e-2, 1e-3) if dtype == torch.float32 else (0.5, 3e-2)
+    assert torch.allclose(ours.weight.grad, ref.weight.grad, atol=gatol,
+                          rtol=grtol)
+    assert torch.allclose(ours.bias.grad, ref.bias.grad, atol=gatol,
+                          rtol=grtol)
+    # running stats must update identically (residual doesn't affect
+    # the reduce pass)
+    assert torch.allclose(ours.running_mean, ref.running_mean, atol=tol,
+                          rtol=tol)
+    assert torch.allclose(ours.running_var, ref.running_var, atol=tol,
+                          rtol=tol)
+
+
+def test_fused_bn_add_relu_eval_mode(dtype=torch.bfloat16):
+    from mgwfbp_amd.kernels.batchnorm import MgxBatchNorm2d
+    C = 128
+    torch.manual_seed(5)
+    ref = nn.BatchNorm2d(C).cuda().eval()
+    ref.running_mean.uniform_(-0.3, 0.3)
+    ref.running_var.uniform_(0.5, 1.5)
+    ours = MgxBatchNorm2d(C).cuda().eval()
+    ours.load_state_dict(ref.state_dict())
+    ours.fuse_relu = True
+    x = _mk(C, dtype=dtype)
+    r = _mk(C, dtype=dtype, seed=7)
+    with torch.no_grad():
+        y_ref = torch.nn.functional.relu(ref(x.float()) + r.float())
+        y = ours.forward_add_relu(x, r)
+    assert torch.allclose(y.float(), y_ref, atol=5e-2, rtol=5e-2)
+
+
+def test_resnet50_bnaddrelu_converted_and_trains():
+    """convert_batchnorm must fuse the BNAddReLU epilogues and the model
+    must step (the path bench.py runs)."""
+    from mgwfbp_amd import models
+    from mgwfbp_amd.kernels.batchnorm import (convert_batchnorm,
+                                              MgxBatchNorm2d)
+    from mgwfbp_amd.models.common import BNAddReLU
+    net = models.resnet50(num_classes=100).cuda().to(
+        memory_format=torch.channels_last)
+    convert_batchnorm(net, fuse_relu=True, only_fused=True)
+    n_fused_add = sum(1 for m in net.modules()
+                      if isinstance(m, BNAddReLU)
+                      and isinstance(m.bn, MgxBatchNorm2d)
+                      and m.bn.fuse_relu)
+    assert n_fused_add == 16   # one per bottleneck
+    opt = torch.optim.SGD(net.parameters(), lr=0.01)
+    x = torch.randn(4, 3, 64, 64, device='cuda').to(
+        memory_format=torch.channels_last)
+    y = torch.randint(0, 100, (4,), device='cuda')
+    with torch.autocast('cuda', dtype=torch.bfloat16):
+        loss = nn.CrossEntropyLoss()(net(x), y)
+    loss.backward()
+    opt.step()
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss)
