@@ -133,9 +133,11 @@ def test_convert_fuses_bnrelu_units_and_sequentials():
                 if isinstance(m, MgxBatchNorm2d) and m.fuse_relu)
     plain = sum(1 for m in net.modules()
                 if isinstance(m, MgxBatchNorm2d) and not m.fuse_relu)
-    # stem + 2 per bottleneck x16 = 33 fused; bn3 x16 + 4 downsample = 20
-    assert fused == 33, fused
-    assert plain == 20, plain
+    # stem + 2 per bottleneck x16 = 33 relu-fused, plus the 16 BNAddReLU
+    # epilogues (bn3, residual-fused) = 49; only the 4 downsample BNs
+    # stay plain
+    assert fused == 49, fused
+    assert plain == 4, plain
     # vgg-style sequential fusion
     vgg = models.vgg16().cuda().to(memory_format=torch.channels_last)
     convert_batchnorm(vgg)
