@@ -80,11 +80,17 @@ def measure_hook_path(n=2000):
     params = {k: opt._named_parameters[k] for k in names}
     # fire in backward order, full passes
     order = list(reversed(names))
+    def reset_flags():
+        # what synchronize() does at step end (flags are per-step)
+        for gi in range(len(opt._groups)):
+            opt._groups_flags[gi] = [0] * len(opt._groups_flags[gi])
+
     t0 = time.time()
     passes = max(1, n // len(order))
     for _ in range(passes):
         for k in order:
             hooks[k](params[k])
+        reset_flags()
     t_hook = (time.time() - t0) / (passes * len(order))
     assert enqueued[0] == passes * len(order)
     return t_hook

@@ -91,10 +91,12 @@ class MgxBatchNorm2d(nn.BatchNorm2d):
                 y = torch.nn.functional.relu(y, inplace=True)
             return y
         if self.training:
-            if self.num_batches_tracked is not None:
-                self.num_batches_tracked.add_(1)
             momentum = self.momentum
             if momentum is None:
+                # cumulative-average mode is the only consumer of the
+                # counter; with a fixed momentum the per-layer
+                # .add_(1) is 53 dead launches per resnet50 step
+                self.num_batches_tracked.add_(1)
                 momentum = 1.0 / float(self.num_batches_tracked)
             return _BatchNormFunc.apply(x, self.weight, self.bias,
                                         self.running_mean,
@@ -120,10 +122,9 @@ class MgxBatchNorm2d(nn.BatchNorm2d):
         residual = residual.contiguous(
             memory_format=torch.channels_last)
         if self.training:
-            if self.num_batches_tracked is not None:
-                self.num_batches_tracked.add_(1)
             momentum = self.momentum
             if momentum is None:
+                self.num_batches_tracked.add_(1)
                 momentum = 1.0 / float(self.num_batches_tracked)
             return _BatchNormAddReLUFunc.apply(
                 x, residual, self.weight, self.bias, self.running_mean,
