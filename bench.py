@@ -26,8 +26,8 @@ import torch
 def main():
     parser = argparse.ArgumentParser()
     parser.add_argument('--gpus', type=int, default=1)
-    parser.add_argument('--steps', type=int, default=40)
-    parser.add_argument('--warmup', type=int, default=10)
+    parser.add_argument('--steps', type=int, default=60)
+    parser.add_argument('--warmup', type=int, default=15)
     parser.add_argument('--model', type=str, default='resnet50')
     parser.add_argument('--batch-size', type=int, default=128)
     parser.add_argument('--dataset', type=str, default='imagenet')

@@ -142,17 +142,37 @@ class AudioDataLoader(DataLoader):
         super().__init__(dataset, collate_fn=an4_collate, **kwargs)
 
 
-def create_manifest(data_path, manifest_path):
-    """Write "wav,txt" manifest rows sorted by file size (a proxy for
-    duration; reference audio_data/utils.py:11-37 used sox)."""
+def wav_duration(path):
+    """Duration in seconds from the RIFF header (stdlib wave — the
+    reference shells out to sox for this, audio_data/utils.py:24-27)."""
+    import wave
+    with wave.open(path, 'rb') as w:
+        rate = w.getframerate()
+        return w.getnframes() / float(rate) if rate else 0.0
+
+
+def create_manifest(data_path, manifest_path, min_duration=None,
+                    max_duration=None):
+    """Write "wav,txt" manifest rows sorted by duration, optionally
+    dropping clips outside [min_duration, max_duration] seconds
+    (reference audio_data/utils.py:11-37, sox-free)."""
     pairs = []
     for root, _, files in os.walk(data_path):
         for fn in files:
             if fn.endswith('.wav'):
                 wav = os.path.join(root, fn)
                 txt = wav.replace('/wav/', '/txt/').replace('.wav', '.txt')
-                if os.path.exists(txt):
-                    pairs.append((os.path.getsize(wav), wav, txt))
+                if not os.path.exists(txt):
+                    continue
+                try:
+                    dur = wav_duration(wav)
+                except Exception:
+                    dur = os.path.getsize(wav) / 32000.0
+                if min_duration is not None and dur < min_duration:
+                    continue
+                if max_duration is not None and dur > max_duration:
+                    continue
+                pairs.append((dur, wav, txt))
     pairs.sort()
     with open(manifest_path, 'w') as f:
         for _, wav, txt in pairs:
