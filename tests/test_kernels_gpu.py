@@ -319,3 +319,32 @@ class TestCommCoreBindings:
         torch.cuda.synchronize()
         ref = TopKCompressor.decompress((values, indices), numel)
         assert torch.allclose(rebuilt, ref)
+
+
+class TestClipMerged:
+    def test_clip_merged_lstman4_shapes(self):
+        """The optimizer's internal device-side clip on an LSTM-AN4-like
+        parameter set (many small RNN grads — the alpha-dominated model
+        whose entry passes norm_clip=400)."""
+        import torch.nn as nn
+        from mgwfbp_amd.distributed_optimizer import DistributedOptimizer
+        torch.manual_seed(0)
+        net = nn.ModuleList([nn.LSTM(96, 192, batch_first=True),
+                             nn.Linear(192, 29)]).cuda()
+        opt = DistributedOptimizer(
+            torch.optim.SGD(net.parameters(), lr=1e-4),
+            named_parameters=list(net.named_parameters()),
+            norm_clip=400, threshold=1 << 30)
+        key = opt._group_keys[0]
+        flat = opt._merged_parameters[key]
+        flat.normal_(0, 10.0)   # force a big norm
+        before = float(flat.norm(2))
+        opt._clip_merged(flat)
+        torch.cuda.synchronize()
+        after = float(flat.norm(2))
+        bound = (1.0 / 1) ** 0.5 * 400
+        assert before > bound
+        assert after <= bound * (1 + 1e-4)
+        # per-param grad views see the clipped values (zero-copy arena)
+        g = torch.cat([p.grad.reshape(-1) for p in net.parameters()])
+        assert abs(float(g.norm(2)) - after) / after < 1e-3
