@@ -286,6 +286,193 @@ __global__ __launch_bounds__(kBlock) void bn_fwd_finalize_kernel(
   }
 }
 
+// ------------------------------------- fused finalize (small layers) ----
+// For small inputs (rblocks <= kFusedFinBlocks, C <= 2048 so one block's
+// octet tiling covers all channels) the separate finalize kernel is pure
+// launch floor (~5 us for reading a few KB — resnet20's graph-replayed
+// step is ~560 such launches). Each norm/dx block re-derives the channel
+// sums from the partials directly in its per-octet preamble (<= 64x16
+// L2-hot loads per octet) and block 0 writes the stat tensors; the
+// finalize launch disappears. 6 -> 4 kernels per BN layer fwd+bwd.
+
+constexpr int kFusedFinBlocks = 64;
+
+__device__ inline void partial_sums8(const float* __restrict__ partial,
+                                     long nblk, long C, long c0,
+                                     float s[8], float q[8]) {
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    s[i] = 0.f;
+    q[i] = 0.f;
+  }
+  for (long b = 0; b < nblk; ++b) {
+    const float* row = partial + b * 2 * C;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      s[i] += row[c0 + i];
+      q[i] += row[C + c0 + i];
+    }
+  }
+}
+
+template <typename T, bool ADD>
+__global__ __launch_bounds__(kBlock) void bn_fwd_norm_fin_kernel(
+    const T* __restrict__ x, const T* __restrict__ res, T* __restrict__ y,
+    long rows, long C, const float* __restrict__ partial, long nblk,
+    float M, float eps, float momentum, const float* __restrict__ gamma,
+    const float* __restrict__ beta, float* __restrict__ mean_out,
+    float* __restrict__ invstd_out, float* __restrict__ running_mean,
+    float* __restrict__ running_var, bool relu) {
+  int opr, rpb, o, rl; bool active;
+  tile_map(C, opr, rpb, o, rl, active);
+  if (!active) return;
+  const long octets = C / 8;
+  for (long ob = o; ob < octets; ob += opr) {
+    const long c0 = ob * 8;
+    float s[8], q[8], mu[8], is[8], sc[8], sh[8];
+    partial_sums8(partial, nblk, C, c0, s, q);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      mu[i] = s[i] / M;
+      float var = q[i] / M - mu[i] * mu[i];
+      var = var < 0.f ? 0.f : var;
+      is[i] = rsqrtf(var + eps);
+      const float g = gamma ? gamma[c0 + i] : 1.f;
+      sc[i] = g * is[i];
+      sh[i] = (beta ? beta[c0 + i] : 0.f) - mu[i] * sc[i];
+      if (blockIdx.x == 0 && rl == 0) {
+        mean_out[c0 + i] = mu[i];
+        invstd_out[c0 + i] = is[i];
+        if (running_mean != nullptr) {
+          running_mean[c0 + i] = (1.f - momentum) * running_mean[c0 + i]
+                                 + momentum * mu[i];
+          const float unbiased = M > 1.f ? var * M / (M - 1.f) : var;
+          running_var[c0 + i] = (1.f - momentum) * running_var[c0 + i]
+                                + momentum * unbiased;
+        }
+      }
+    }
+    const long rstride = (long)gridDim.x * rpb;
+    long r = (long)blockIdx.x * rpb + rl;
+    for (; r + 3 * rstride < rows; r += 4 * rstride) {
+      float v[4][8], a[4][8];
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+        VecIO<T>::load(x + (r + u * rstride) * C + c0, v[u]);
+      if (ADD) {
+#pragma unroll
+        for (int u = 0; u < 4; ++u)
+          VecIO<T>::load(res + (r + u * rstride) * C + c0, a[u]);
+      }
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          float t = v[u][i] * sc[i] + sh[i];
+          if (ADD) t += a[u][i];
+          if (relu) t = t > 0.f ? t : 0.f;
+          v[u][i] = t;
+        }
+      }
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+        VecIO<T>::store(y + (r + u * rstride) * C + c0, v[u]);
+    }
+    for (; r < rows; r += rstride) {
+      float v[8], a[8];
+      VecIO<T>::load(x + r * C + c0, v);
+      if (ADD) VecIO<T>::load(res + r * C + c0, a);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        v[i] = v[i] * sc[i] + sh[i];
+        if (ADD) v[i] += a[i];
+        if (relu) v[i] = v[i] > 0.f ? v[i] : 0.f;
+      }
+      VecIO<T>::store(y + r * C + c0, v);
+    }
+  }
+}
+
+template <typename T, bool RELU, bool ADD>
+__global__ __launch_bounds__(kBlock) void bn_bwd_dx_fin_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x,
+    const T* __restrict__ res, T* __restrict__ dx, T* __restrict__ dres,
+    long rows, long C, const float* __restrict__ mean,
+    const float* __restrict__ invstd, const float* __restrict__ gamma,
+    const float* __restrict__ beta, const float* __restrict__ partial,
+    long nblk, float invM, float* __restrict__ dbeta_out,
+    float* __restrict__ dgamma_out) {
+  int opr, rpb, o, rl; bool active;
+  tile_map(C, opr, rpb, o, rl, active);
+  if (!active) return;
+  const long octets = C / 8;
+  for (long ob = o; ob < octets; ob += opr) {
+    const long c0 = ob * 8;
+    float sd[8], sx[8], mu[8], is[8], gi[8], md[8], mx[8], ga[8], be[8];
+    partial_sums8(partial, nblk, C, c0, sd, sx);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      mu[i] = mean[c0 + i];
+      is[i] = invstd[c0 + i];
+      ga[i] = gamma ? gamma[c0 + i] : 1.f;
+      be[i] = beta ? beta[c0 + i] : 0.f;
+      gi[i] = ga[i] * is[i];
+      md[i] = sd[i] * invM;
+      mx[i] = sx[i] * invM;
+      if (blockIdx.x == 0 && rl == 0) {
+        dbeta_out[c0 + i] = sd[i];
+        dgamma_out[c0 + i] = sx[i];
+      }
+    }
+    const long rstride = (long)gridDim.x * rpb;
+    long r = (long)blockIdx.x * rpb + rl;
+    constexpr int U = ADD ? 2 : 4;
+    for (; r + (U - 1) * rstride < rows; r += U * rstride) {
+      float g[U][8], v[U][8], a[U][8];
+#pragma unroll
+      for (int u = 0; u < U; ++u) {
+        VecIO<T>::load(dy + (r + u * rstride) * C + c0, g[u]);
+        VecIO<T>::load(x + (r + u * rstride) * C + c0, v[u]);
+        if (ADD) VecIO<T>::load(res + (r + u * rstride) * C + c0, a[u]);
+      }
+#pragma unroll
+      for (int u = 0; u < U; ++u) {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          const float xh = (v[u][i] - mu[i]) * is[i];
+          if (RELU) {
+            const float rr = ADD ? a[u][i] : 0.f;
+            if (xh * ga[i] + be[i] + rr <= 0.f) g[u][i] = 0.f;
+          }
+          if (ADD) a[u][i] = g[u][i];
+          g[u][i] = gi[i] * (g[u][i] - md[i] - xh * mx[i]);
+        }
+      }
+#pragma unroll
+      for (int u = 0; u < U; ++u) {
+        if (ADD) VecIO<T>::store(dres + (r + u * rstride) * C + c0, a[u]);
+        VecIO<T>::store(dx + (r + u * rstride) * C + c0, g[u]);
+      }
+    }
+    for (; r < rows; r += rstride) {
+      float g[8], v[8], a[8];
+      VecIO<T>::load(dy + r * C + c0, g);
+      VecIO<T>::load(x + r * C + c0, v);
+      if (ADD) VecIO<T>::load(res + r * C + c0, a);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const float xh = (v[i] - mu[i]) * is[i];
+        if (RELU && xh * ga[i] + be[i] + (ADD ? a[i] : 0.f) <= 0.f)
+          g[i] = 0.f;
+        if (ADD) a[i] = g[i];
+        g[i] = gi[i] * (g[i] - md[i] - xh * mx[i]);
+      }
+      if (ADD) VecIO<T>::store(dres + r * C + c0, a);
+      VecIO<T>::store(dx + r * C + c0, g);
+    }
+  }
+}
+
 // --------------------------------------------------- fwd: norm ----------
 // ADD: y = [relu](bn(x) + res) — the ResNet post-add activation folded
 // into the normalize pass: ONE kernel replaces bn_out store + add read/
@@ -628,22 +815,56 @@ std::vector<torch::Tensor> bn_fwd_train(
                     residual.scalar_type() == x.scalar_type(),
                 "residual must match x (channels_last, shape, dtype)");
   }
+  const bool fused_fin = (rblocks <= kFusedFinBlocks && g.C <= 2048);
   DISPATCH_DT(x.scalar_type(), {
     hipLaunchKernelGGL(bn_fwd_reduce_kernel<dt>, dim3(rblocks),
                        dim3(kBlock), 0, stream,
                        reinterpret_cast<const dt*>(x.data_ptr()), g.rows,
                        g.C, partial.data_ptr<float>());
   });
+  const float* gamma_pf = gamma.defined() ? gamma.data_ptr<float>()
+                                          : nullptr;
+  const float* beta_pf = beta.defined() ? beta.data_ptr<float>()
+                                        : nullptr;
+  float* rm_p = running_mean.defined() ? running_mean.data_ptr<float>()
+                                       : nullptr;
+  float* rv_p = running_var.defined() ? running_var.data_ptr<float>()
+                                      : nullptr;
+  if (fused_fin) {
+    // finalize inlined into the normalize pass (see kFusedFinBlocks)
+    DISPATCH_DT(x.scalar_type(), {
+      if (add)
+        hipLaunchKernelGGL((bn_fwd_norm_fin_kernel<dt, true>),
+                           dim3(eblocks), dim3(kBlock), 0, stream,
+                           reinterpret_cast<const dt*>(x.data_ptr()),
+                           reinterpret_cast<const dt*>(
+                               residual.data_ptr()),
+                           reinterpret_cast<dt*>(y.data_ptr()), g.rows,
+                           g.C, partial.data_ptr<float>(), (long)rblocks,
+                           (float)g.rows, (float)eps, (float)momentum,
+                           gamma_pf, beta_pf, mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), rm_p, rv_p, relu);
+      else
+        hipLaunchKernelGGL((bn_fwd_norm_fin_kernel<dt, false>),
+                           dim3(eblocks), dim3(kBlock), 0, stream,
+                           reinterpret_cast<const dt*>(x.data_ptr()),
+                           nullptr,
+                           reinterpret_cast<dt*>(y.data_ptr()), g.rows,
+                           g.C, partial.data_ptr<float>(), (long)rblocks,
+                           (float)g.rows, (float)eps, (float)momentum,
+                           gamma_pf, beta_pf, mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), rm_p, rv_p, relu);
+    });
+    CHECK_HIP(hipGetLastError());
+    return {y, mean, invstd};
+  }
   auto fin = finalize_tree(partial, rblocks, g.C, stream);
   hipLaunchKernelGGL(bn_fwd_finalize_kernel,
                      dim3((g.C + kFinC - 1) / kFinC), dim3(kBlock), 0,
                      stream, fin.first, fin.second, g.C,
                      (float)g.rows, (float)eps, (float)momentum,
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                     running_mean.defined()
-                         ? running_mean.data_ptr<float>() : nullptr,
-                     running_var.defined()
-                         ? running_var.data_ptr<float>() : nullptr);
+                     rm_p, rv_p);
   DISPATCH_DT(x.scalar_type(), {
     if (add)
       hipLaunchKernelGGL((bn_fwd_norm_kernel<dt, true>), dim3(eblocks),
@@ -652,10 +873,7 @@ std::vector<torch::Tensor> bn_fwd_train(
                          reinterpret_cast<const dt*>(residual.data_ptr()),
                          reinterpret_cast<dt*>(y.data_ptr()), g.rows, g.C,
                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                         gamma.defined() ? gamma.data_ptr<float>()
-                                         : nullptr,
-                         beta.defined() ? beta.data_ptr<float>() : nullptr,
-                         relu);
+                         gamma_pf, beta_pf, relu);
     else
       hipLaunchKernelGGL((bn_fwd_norm_kernel<dt, false>), dim3(eblocks),
                          dim3(kBlock), 0, stream,
@@ -663,10 +881,7 @@ std::vector<torch::Tensor> bn_fwd_train(
                          nullptr,
                          reinterpret_cast<dt*>(y.data_ptr()), g.rows, g.C,
                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                         gamma.defined() ? gamma.data_ptr<float>()
-                                         : nullptr,
-                         beta.defined() ? beta.data_ptr<float>() : nullptr,
-                         relu);
+                         gamma_pf, beta_pf, relu);
   });
   CHECK_HIP(hipGetLastError());
   return {y, mean, invstd};
@@ -769,6 +984,38 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                          invstd.data_ptr<float>(), gamma_p, beta_p,
                          partial.data_ptr<float>());
   });
+  const bool fused_fin = (rblocks <= kFusedFinBlocks && g.C <= 2048);
+  if (fused_fin) {
+    DISPATCH_DT(x.scalar_type(), {
+      const dt* res_p = add
+          ? reinterpret_cast<const dt*>(residual.data_ptr()) : nullptr;
+      dt* dres_p = nullptr;
+      if (add) {
+        dres = torch::empty_like(x);
+        dres_p = reinterpret_cast<dt*>(dres.data_ptr());
+      }
+      auto launch = [&](auto kern) {
+        hipLaunchKernelGGL(kern, dim3(eblocks), dim3(kBlock), 0, stream,
+                           reinterpret_cast<const dt*>(dy.data_ptr()),
+                           reinterpret_cast<const dt*>(x.data_ptr()),
+                           res_p, reinterpret_cast<dt*>(dx.data_ptr()),
+                           dres_p, g.rows, g.C, mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), gamma_p, beta_p,
+                           partial.data_ptr<float>(), (long)rblocks,
+                           1.f / (float)g.rows, dbeta.data_ptr<float>(),
+                           dgamma.data_ptr<float>());
+      };
+      if (add)
+        launch(bn_bwd_dx_fin_kernel<dt, true, true>);
+      else if (relu)
+        launch(bn_bwd_dx_fin_kernel<dt, true, false>);
+      else
+        launch(bn_bwd_dx_fin_kernel<dt, false, false>);
+    });
+    CHECK_HIP(hipGetLastError());
+    if (add) return {dx, dgamma, dbeta, dres};
+    return {dx, dgamma, dbeta};
+  }
   auto finb = finalize_tree(partial, rblocks, g.C, stream);
   hipLaunchKernelGGL(bn_bwd_finalize_kernel,
                      dim3((g.C + kFinC - 1) / kFinC), dim3(kBlock), 0,
