@@ -297,6 +297,16 @@ __global__ __launch_bounds__(kBlock) void bn_fwd_finalize_kernel(
 
 constexpr int kFusedFinBlocks = 64;
 
+// runtime A/B switch (same-box measurement): MGX_BN_FUSED_FIN=0 keeps
+// the separate finalize kernels everywhere
+inline bool fused_fin_enabled() {
+  static const bool on = [] {
+    const char* v = getenv("MGX_BN_FUSED_FIN");
+    return v == nullptr || v[0] != '0';
+  }();
+  return on;
+}
+
 __device__ inline void partial_sums8(const float* __restrict__ partial,
                                      long nblk, long C, long c0,
                                      float s[8], float q[8]) {
@@ -835,7 +845,8 @@ std::vector<torch::Tensor> bn_fwd_train(
                     residual.scalar_type() == x.scalar_type(),
                 "residual must match x (channels_last, shape, dtype)");
   }
-  const bool fused_fin = (rblocks <= kFusedFinBlocks && g.C <= 2048);
+  const bool fused_fin = (fused_fin_enabled() &&
+                          rblocks <= kFusedFinBlocks && g.C <= 2048);
   DISPATCH_DT(x.scalar_type(), {
     hipLaunchKernelGGL(bn_fwd_reduce_kernel<dt>, dim3(rblocks),
                        dim3(kBlock), 0, stream,
@@ -1004,7 +1015,8 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                          invstd.data_ptr<float>(), gamma_p, beta_p,
                          partial.data_ptr<float>());
   });
-  const bool fused_fin = (rblocks <= kFusedFinBlocks && g.C <= 2048);
+  const bool fused_fin = (fused_fin_enabled() &&
+                          rblocks <= kFusedFinBlocks && g.C <= 2048);
   if (fused_fin) {
     DISPATCH_DT(x.scalar_type(), {
       const dt* res_p = add
