@@ -307,22 +307,50 @@ inline bool fused_fin_enabled() {
   return on;
 }
 
-__device__ inline void partial_sums8(const float* __restrict__ partial,
-                                     long nblk, long C, long c0,
-                                     float s[8], float q[8]) {
+// Block-cooperative finalize of the per-octet channel sums: the rpb
+// row-threads of each octet column split the nblk partial rows, then an
+// LDS tree reduces to the rl==0 leader (a serial leader loop measured
+// 9% SLOWER end-to-end on resnet20 — every block stalled ~10us in the
+// preamble). Returns true on the leader lane with (s, q) final.
+__device__ inline bool partial_sums_coop(
+    const float* __restrict__ partial, long nblk, long C, long c0,
+    int opr, int rpb, int o, int rl, bool active,
+    float* __restrict__ lds, float s[8], float q[8]) {
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
     s[i] = 0.f;
     q[i] = 0.f;
   }
-  for (long b = 0; b < nblk; ++b) {
-    const float* row = partial + b * 2 * C;
+  if (active) {
+    for (long b = rl; b < nblk; b += rpb) {
+      const float* row = partial + b * 2 * C;
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      s[i] += row[c0 + i];
-      q[i] += row[C + c0 + i];
+      for (int i = 0; i < 8; ++i) {
+        s[i] += row[c0 + i];
+        q[i] += row[C + c0 + i];
+      }
     }
   }
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    lds[(rl * opr + o) * 16 + i] = active ? s[i] : 0.f;
+    lds[(rl * opr + o) * 16 + 8 + i] = active ? q[i] : 0.f;
+  }
+  __syncthreads();
+  if (rl != 0 || !active) return false;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    s[i] = 0.f;
+    q[i] = 0.f;
+  }
+  for (int r = 0; r < rpb; ++r) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      s[i] += lds[(r * opr + o) * 16 + i];
+      q[i] += lds[(r * opr + o) * 16 + 8 + i];
+    }
+  }
+  return true;
 }
 
 // Host guarantees opr == octets here (C <= 2048), so every thread takes
@@ -342,27 +370,31 @@ __global__ __launch_bounds__(kBlock) void bn_fwd_norm_fin_kernel(
   int opr, rpb, o, rl; bool active;
   tile_map(C, opr, rpb, o, rl, active);
   const long c0 = (long)o * 8;
+  __shared__ float lds[kBlock * 16];
   __shared__ float smu[kBlock * 8], sis[kBlock * 8];   // [octet*8+i]
-  if (active && rl == 0) {
+  {
     float s[8], q[8];
-    partial_sums8(partial, nblk, C, c0, s, q);
+    const bool leader = partial_sums_coop(partial, nblk, C, c0, opr, rpb,
+                                          o, rl, active, lds, s, q);
+    if (leader) {
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      const float m = s[i] / M;
-      float var = q[i] / M - m * m;
-      var = var < 0.f ? 0.f : var;
-      const float is = rsqrtf(var + eps);
-      smu[o * 8 + i] = m;
-      sis[o * 8 + i] = is;
-      if (blockIdx.x == 0) {
-        mean_out[c0 + i] = m;
-        invstd_out[c0 + i] = is;
-        if (running_mean != nullptr) {
-          running_mean[c0 + i] = (1.f - momentum) * running_mean[c0 + i]
-                                 + momentum * m;
-          const float unbiased = M > 1.f ? var * M / (M - 1.f) : var;
-          running_var[c0 + i] = (1.f - momentum) * running_var[c0 + i]
-                                + momentum * unbiased;
+      for (int i = 0; i < 8; ++i) {
+        const float m = s[i] / M;
+        float var = q[i] / M - m * m;
+        var = var < 0.f ? 0.f : var;
+        const float is = rsqrtf(var + eps);
+        smu[o * 8 + i] = m;
+        sis[o * 8 + i] = is;
+        if (blockIdx.x == 0) {
+          mean_out[c0 + i] = m;
+          invstd_out[c0 + i] = is;
+          if (running_mean != nullptr) {
+            running_mean[c0 + i] =
+                (1.f - momentum) * running_mean[c0 + i] + momentum * m;
+            const float unbiased = M > 1.f ? var * M / (M - 1.f) : var;
+            running_var[c0 + i] = (1.f - momentum) * running_var[c0 + i]
+                                  + momentum * unbiased;
+          }
         }
       }
     }
@@ -428,17 +460,21 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_dx_fin_kernel(
   int opr, rpb, o, rl; bool active;
   tile_map(C, opr, rpb, o, rl, active);
   const long c0 = (long)o * 8;
+  __shared__ float lds[kBlock * 16];
   __shared__ float ssd[kBlock * 8], ssx[kBlock * 8];   // [octet*8+i]
-  if (active && rl == 0) {
+  {
     float sd[8], sx[8];
-    partial_sums8(partial, nblk, C, c0, sd, sx);
+    const bool leader = partial_sums_coop(partial, nblk, C, c0, opr, rpb,
+                                          o, rl, active, lds, sd, sx);
+    if (leader) {
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      ssd[o * 8 + i] = sd[i];
-      ssx[o * 8 + i] = sx[i];
-      if (blockIdx.x == 0) {
-        dbeta_out[c0 + i] = sd[i];
-        dgamma_out[c0 + i] = sx[i];
+      for (int i = 0; i < 8; ++i) {
+        ssd[o * 8 + i] = sd[i];
+        ssx[o * 8 + i] = sx[i];
+        if (blockIdx.x == 0) {
+          dbeta_out[c0 + i] = sd[i];
+          dgamma_out[c0 + i] = sx[i];
+        }
       }
     }
   }
