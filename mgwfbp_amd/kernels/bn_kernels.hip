@@ -297,12 +297,17 @@ __global__ __launch_bounds__(kBlock) void bn_fwd_finalize_kernel(
 
 constexpr int kFusedFinBlocks = 64;
 
-// runtime A/B switch (same-box measurement): MGX_BN_FUSED_FIN=0 keeps
-// the separate finalize kernels everywhere
+// Runtime switch, DEFAULT OFF: same-box A/B (resnet20 bs32, 300-step
+// graph replay, 2026-09-14) measured the fused-finalize variant 9%
+// SLOWER end-to-end (1.547 vs 1.412 ms/step) both with a serial leader
+// preamble and with the cooperative LDS-tree one — the 32 KB LDS
+// footprint + preamble barrier cost the streaming loop more than the
+// ~4.6 us finalize launch it saves. Kept behind MGX_BN_FUSED_FIN=1 for
+// re-evaluation on future ROCm versions.
 inline bool fused_fin_enabled() {
   static const bool on = [] {
     const char* v = getenv("MGX_BN_FUSED_FIN");
-    return v == nullptr || v[0] != '0';
+    return v != nullptr && v[0] == '1';
   }();
   return on;
 }
