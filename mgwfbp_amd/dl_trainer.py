@@ -172,6 +172,16 @@ class DLTrainer:
                                           only_fused=(mode == 'relu'))
                 except Exception as e:
                     logger.warning('fused BN unavailable: %s', e)
+            # gather-based NHWC maxpool (no atomics in backward):
+            # 440 us -> ~80 us per VGG pool, 311 -> ~60 us resnet stem
+            if os.environ.get('MGX_FUSED_POOL', '1') == '1':
+                try:
+                    from .kernels.pooling import convert_maxpool
+                    from .kernels import hip_available
+                    if hip_available():
+                        convert_maxpool(self.net)
+                except Exception as e:
+                    logger.warning('fused maxpool unavailable: %s', e)
         if settings.DEBUG and rank == 0:
             logger.info('%s: %d parameters', dnn,
                         sum(p.numel() for p in self.net.parameters()))

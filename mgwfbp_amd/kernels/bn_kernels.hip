@@ -795,6 +795,119 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_dx_kernel(
   }
 }
 
+// ----------------------------------------------- NHWC max pooling -------
+// Hand-written replacement for at::native::max_pool_{forward,backward}_
+// nhwc: torch's NHWC maxpool backward scatters through ATOMICS and
+// measured 440 us/call on VGG-16's 5 pools (11% of the step) and
+// 311 us on resnet50's stem pool. Here:
+//   fwd:  per output octet, scan the KxK window, store max + a uint8
+//         ARGMAX (window-local 0..K*K-1) — same thread tiling as BN.
+//   bwd:  GATHER per input octet: enumerate the <= ceil(K/S)^2 windows
+//         that contain this input pixel, read their dy+idx octets, and
+//         accumulate where the argmax points back here. No atomics, no
+//         full-tensor zero-fill, deterministic.
+// Gradient ties follow the saved argmax (self-consistent; torch
+// likewise routes the gradient to its own saved index).
+
+template <typename T>
+__global__ __launch_bounds__(kBlock) void maxpool_fwd_kernel(
+    const T* __restrict__ x, T* __restrict__ y,
+    unsigned char* __restrict__ idx, long N, long C, long H, long W,
+    long OH, long OW, int K, int S, int P) {
+  int opr, rpb, o, rl; bool active;
+  tile_map(C, opr, rpb, o, rl, active);
+  if (!active) return;
+  const long octets = C / 8;
+  const long orows = N * OH * OW;
+  const long rstride = (long)gridDim.x * rpb;
+  for (long ob = o; ob < octets; ob += opr) {
+    const long c0 = ob * 8;
+    for (long r = (long)blockIdx.x * rpb + rl; r < orows; r += rstride) {
+      const long ow = r % OW;
+      const long oh = (r / OW) % OH;
+      const long n = r / (OW * OH);
+      const long ih0 = oh * S - P;
+      const long iw0 = ow * S - P;
+      float best[8];
+      unsigned char bi[8];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        best[i] = -3.4e38f;
+        bi[i] = 0;
+      }
+      for (int kh = 0; kh < K; ++kh) {
+        const long ih = ih0 + kh;
+        if (ih < 0 || ih >= H) continue;
+        for (int kw = 0; kw < K; ++kw) {
+          const long iw = iw0 + kw;
+          if (iw < 0 || iw >= W) continue;
+          float v[8];
+          VecIO<T>::load(x + ((n * H + ih) * W + iw) * C + c0, v);
+          const unsigned char li = (unsigned char)(kh * K + kw);
+#pragma unroll
+          for (int i = 0; i < 8; ++i) {
+            if (v[i] > best[i]) {
+              best[i] = v[i];
+              bi[i] = li;
+            }
+          }
+        }
+      }
+      VecIO<T>::store(y + r * C + c0, best);
+      // 8 uint8 argmaxes as one 8-byte store
+      uint2 packed;
+      unsigned char* pb = reinterpret_cast<unsigned char*>(&packed);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) pb[i] = bi[i];
+      *reinterpret_cast<uint2*>(idx + r * C + c0) = packed;
+    }
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(kBlock) void maxpool_bwd_kernel(
+    const T* __restrict__ dy, const unsigned char* __restrict__ idx,
+    T* __restrict__ dx, long N, long C, long H, long W, long OH, long OW,
+    int K, int S, int P) {
+  int opr, rpb, o, rl; bool active;
+  tile_map(C, opr, rpb, o, rl, active);
+  if (!active) return;
+  const long octets = C / 8;
+  const long irows = N * H * W;
+  const long rstride = (long)gridDim.x * rpb;
+  for (long ob = o; ob < octets; ob += opr) {
+    const long c0 = ob * 8;
+    for (long r = (long)blockIdx.x * rpb + rl; r < irows; r += rstride) {
+      const long iw = r % W;
+      const long ih = (r / W) % H;
+      const long n = r / (W * H);
+      float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      // windows (oh, ow) with oh*S - P <= ih < oh*S - P + K
+      const long oh_hi = (ih + P) / S;
+      const long oh_lo = (ih + P - K + S) / S;   // ceil((ih+P-K+1)/S)
+      const long ow_hi = (iw + P) / S;
+      const long ow_lo = (iw + P - K + S) / S;
+      for (long oh = max(oh_lo, 0L); oh <= min(oh_hi, OH - 1); ++oh) {
+        for (long ow = max(ow_lo, 0L); ow <= min(ow_hi, OW - 1); ++ow) {
+          const long orow = (n * OH + oh) * OW + ow;
+          const unsigned char li = (unsigned char)(
+              (ih - (oh * S - P)) * K + (iw - (ow * S - P)));
+          const uint2 packed =
+              *reinterpret_cast<const uint2*>(idx + orow * C + c0);
+          const unsigned char* pb =
+              reinterpret_cast<const unsigned char*>(&packed);
+          float g[8];
+          VecIO<T>::load(dy + orow * C + c0, g);
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+            if (pb[i] == li) acc[i] += g[i];
+        }
+      }
+      VecIO<T>::store(dx + r * C + c0, acc);
+    }
+  }
+}
+
 // ------------------------------------------------- host plumbing --------
 
 struct Geometry {
@@ -1136,9 +1249,69 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   return {dx, dgamma, dbeta};
 }
 
+std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x, long K, long S,
+                                       long P) {
+  auto g = geom(x);
+  const long N = x.size(0), H = x.size(2), W = x.size(3);
+  const long OH = (H + 2 * P - K) / S + 1;
+  const long OW = (W + 2 * P - K) / S + 1;
+  TORCH_CHECK(K * K <= 255, "argmax stored as uint8");
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  auto y = torch::empty(
+      {N, g.C, OH, OW},
+      x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  auto idx = torch::empty(
+      {N * OH * OW * g.C},
+      torch::TensorOptions().dtype(torch::kUInt8).device(x.device()));
+  const long orows = N * OH * OW;
+  long nblk = (orows + g.rpb - 1) / g.rpb;
+  if (nblk > 2048) nblk = 2048;
+  DISPATCH_DT(x.scalar_type(), {
+    hipLaunchKernelGGL(maxpool_fwd_kernel<dt>,
+                       dim3((int)std::max(nblk, 1L)), dim3(kBlock), 0,
+                       stream, reinterpret_cast<const dt*>(x.data_ptr()),
+                       reinterpret_cast<dt*>(y.data_ptr()),
+                       idx.data_ptr<unsigned char>(), N, g.C, H, W, OH,
+                       OW, (int)K, (int)S, (int)P);
+  });
+  CHECK_HIP(hipGetLastError());
+  return {y, idx};
+}
+
+torch::Tensor maxpool_bwd(torch::Tensor dy, torch::Tensor idx, long N,
+                          long C, long H, long W, long K, long S, long P) {
+  auto gdy = geom(dy);
+  const long OH = dy.size(2), OW = dy.size(3);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  auto dx = torch::empty(
+      {N, C, H, W},
+      dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+  const long irows = N * H * W;
+  const int opr = (int)std::min(C / 8, (long)kBlock);
+  const int rpb = kBlock / opr;
+  long nblk = (irows + rpb - 1) / rpb;
+  if (nblk > 2048) nblk = 2048;
+  DISPATCH_DT(dy.scalar_type(), {
+    hipLaunchKernelGGL(maxpool_bwd_kernel<dt>,
+                       dim3((int)std::max(nblk, 1L)), dim3(kBlock), 0,
+                       stream,
+                       reinterpret_cast<const dt*>(dy.data_ptr()),
+                       idx.data_ptr<unsigned char>(),
+                       reinterpret_cast<dt*>(dx.data_ptr()), N, C, H, W,
+                       OH, OW, (int)K, (int)S, (int)P);
+  });
+  CHECK_HIP(hipGetLastError());
+  (void)gdy;
+  return dx;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("maxpool_fwd", &maxpool_fwd,
+        "NHWC max pool forward; returns (y, uint8 argmax)");
+  m.def("maxpool_bwd", &maxpool_bwd,
+        "NHWC max pool backward (gather, no atomics)");
   m.def("bn_fwd_train", &bn_fwd_train,
         "NHWC BN training forward (optionally y=relu(bn(x)+residual)); "
         "returns (y, mean, invstd)",
