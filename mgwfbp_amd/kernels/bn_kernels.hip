@@ -719,7 +719,7 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_finalize_kernel(
 // ADD: dy is first gated by the recomputed post-add sign; the gated dy
 // IS the residual branch's gradient, stored to dres in the same pass
 // (saving the separate relu-backward kernel of the unfused chain).
-template <typename T, bool RELU, bool ADD>
+template <typename T, bool RELU, bool ADD, int UROWS = (ADD ? 2 : 4)>
 __global__ __launch_bounds__(kBlock) void bn_bwd_dx_kernel(
     const T* __restrict__ dy, const T* __restrict__ x,
     const T* __restrict__ res, T* __restrict__ dx, T* __restrict__ dres,
@@ -746,9 +746,10 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_dx_kernel(
     }
     const long rstride = (long)gridDim.x * rpb;
     long r = (long)blockIdx.x * rpb + rl;
-    // U rows in flight: 4 for the 2-input case (dy+x = 8 loads), 2 when
-    // the residual makes it 3 inputs + 2 outputs (VGPR pressure)
-    constexpr int U = ADD ? 2 : 4;
+    // UROWS rows in flight: 4 for the 2-input case (dy+x = 8 loads);
+    // the 3-input residual case defaults to 2, with 3 selectable via
+    // MGX_BN_DX_U3 for a same-box VGPR-vs-latency A/B
+    constexpr int U = UROWS;
     for (; r + (U - 1) * rstride < rows; r += U * rstride) {
       float g[U][8], v[U][8], a[U][8];
 #pragma unroll
@@ -1207,8 +1208,25 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                      dim3((g.C + kFinC - 1) / kFinC), dim3(kBlock), 0,
                      stream, finb.first, finb.second, g.C,
                      dbeta.data_ptr<float>(), dgamma.data_ptr<float>());
+  static const bool dx_u3 = [] {
+    const char* v = getenv("MGX_BN_DX_U3");
+    return v != nullptr && v[0] == '1';
+  }();
   DISPATCH_DT(x.scalar_type(), {
-    if (add) {
+    if (add && dx_u3) {
+      dres = torch::empty_like(x);
+      hipLaunchKernelGGL((bn_bwd_dx_kernel<dt, true, true, 3>),
+                         dim3(eblocks), dim3(kBlock), 0, stream,
+                         reinterpret_cast<const dt*>(dy.data_ptr()),
+                         reinterpret_cast<const dt*>(x.data_ptr()),
+                         reinterpret_cast<const dt*>(residual.data_ptr()),
+                         reinterpret_cast<dt*>(dx.data_ptr()),
+                         reinterpret_cast<dt*>(dres.data_ptr()),
+                         g.rows, g.C,
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         gamma_p, beta_p, dbeta.data_ptr<float>(),
+                         dgamma.data_ptr<float>(), 1.f / (float)g.rows);
+    } else if (add) {
       dres = torch::empty_like(x);
       hipLaunchKernelGGL((bn_bwd_dx_kernel<dt, true, true>),
                          dim3(eblocks), dim3(kBlock), 0, stream,
