@@ -648,28 +648,30 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_reduce_kernel(
       }
       const long rstride = (long)gridDim.x * rpb;
       long r = (long)blockIdx.x * rpb + rl;
-      for (; r + rstride < rows; r += 2 * rstride) {  // 4 loads in flight
-        float g0[8], v0[8], g1[8], v1[8], a0[8], a1[8];
-        VecIO<T>::load(dy + r * C + c0, g0);
-        VecIO<T>::load(x + r * C + c0, v0);
-        VecIO<T>::load(dy + (r + rstride) * C + c0, g1);
-        VecIO<T>::load(x + (r + rstride) * C + c0, v1);
-        if (ADD) {
-          VecIO<T>::load(res + r * C + c0, a0);
-          VecIO<T>::load(res + (r + rstride) * C + c0, a1);
+      // U rows in flight (2U or 3U 16 B loads) — matched to the dx
+      // kernel's measured sweet spot (3 rows for the 3-input case)
+      constexpr int U = ADD ? 3 : 4;
+      for (; r + (U - 1) * rstride < rows; r += U * rstride) {
+        float g[U][8], v[U][8], a[U][8];
+#pragma unroll
+        for (int u = 0; u < U; ++u) {
+          VecIO<T>::load(dy + (r + u * rstride) * C + c0, g[u]);
+          VecIO<T>::load(x + (r + u * rstride) * C + c0, v[u]);
+          if (ADD)
+            VecIO<T>::load(res + (r + u * rstride) * C + c0, a[u]);
         }
 #pragma unroll
-        for (int i = 0; i < 8; ++i) {
-          const float xh0 = (v0[i] - mu[i]) * is[i];
-          const float xh1 = (v1[i] - mu[i]) * is[i];
-          if (RELU) {
-            const float r0 = ADD ? a0[i] : 0.f;
-            const float r1 = ADD ? a1[i] : 0.f;
-            if (xh0 * ga[i] + be[i] + r0 <= 0.f) g0[i] = 0.f;
-            if (xh1 * ga[i] + be[i] + r1 <= 0.f) g1[i] = 0.f;
+        for (int u = 0; u < U; ++u) {
+#pragma unroll
+          for (int i = 0; i < 8; ++i) {
+            const float xh = (v[u][i] - mu[i]) * is[i];
+            if (RELU) {
+              const float rr = ADD ? a[u][i] : 0.f;
+              if (xh * ga[i] + be[i] + rr <= 0.f) g[u][i] = 0.f;
+            }
+            sd[i] += g[u][i];
+            sx[i] += g[u][i] * xh;
           }
-          sd[i] += g0[i] + g1[i];
-          sx[i] += g0[i] * xh0 + g1[i] * xh1;
         }
       }
       for (; r < rows; r += rstride) {
