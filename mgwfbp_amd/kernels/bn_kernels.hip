@@ -549,7 +549,7 @@ __global__ __launch_bounds__(kBlock) void bn_bwd_dx_fin_kernel(
 // into the normalize pass: ONE kernel replaces bn_out store + add read/
 // read/write + relu read/write (4 extra full-tensor HBM passes in the
 // unfused eager chain). res is loaded once per element alongside x.
-template <typename T, bool ADD>
+template <typename T, bool ADD, int UROWS = 4>
 __global__ __launch_bounds__(kBlock) void bn_fwd_norm_kernel(
     const T* __restrict__ x, const T* __restrict__ res, T* __restrict__ y,
     long rows, long C, const float* __restrict__ mean,
@@ -574,18 +574,19 @@ __global__ __launch_bounds__(kBlock) void bn_fwd_norm_kernel(
     // flight per thread — the 2x version measured only 2.6 TB/s on the
     // C=64 layers (latency-bound; the 4x-unrolled reduce kernel on the
     // same shapes sustains materially more)
-    for (; r + 3 * rstride < rows; r += 4 * rstride) {
-      float v[4][8], a[4][8];
+    constexpr int U = UROWS;
+    for (; r + (U - 1) * rstride < rows; r += U * rstride) {
+      float v[U][8], a[U][8];
 #pragma unroll
-      for (int u = 0; u < 4; ++u)
+      for (int u = 0; u < U; ++u)
         VecIO<T>::load(x + (r + u * rstride) * C + c0, v[u]);
       if (ADD) {
 #pragma unroll
-        for (int u = 0; u < 4; ++u)
+        for (int u = 0; u < U; ++u)
           VecIO<T>::load(res + (r + u * rstride) * C + c0, a[u]);
       }
 #pragma unroll
-      for (int u = 0; u < 4; ++u) {
+      for (int u = 0; u < U; ++u) {
 #pragma unroll
         for (int i = 0; i < 8; ++i) {
           float t = v[u][i] * sc[i] + sh[i];
@@ -595,7 +596,7 @@ __global__ __launch_bounds__(kBlock) void bn_fwd_norm_kernel(
         }
       }
 #pragma unroll
-      for (int u = 0; u < 4; ++u)
+      for (int u = 0; u < U; ++u)
         VecIO<T>::store(y + (r + u * rstride) * C + c0, v[u]);
     }
     for (; r < rows; r += rstride) {
@@ -1054,7 +1055,19 @@ std::vector<torch::Tensor> bn_fwd_train(
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      rm_p, rv_p);
   DISPATCH_DT(x.scalar_type(), {
-    if (add)
+    static const bool norm_u3 = [] {
+      const char* v = getenv("MGX_BN_NORM_U3");
+      return v != nullptr && v[0] == '1';
+    }();
+    if (add && norm_u3)
+      hipLaunchKernelGGL((bn_fwd_norm_kernel<dt, true, 3>),
+                         dim3(eblocks), dim3(kBlock), 0, stream,
+                         reinterpret_cast<const dt*>(x.data_ptr()),
+                         reinterpret_cast<const dt*>(residual.data_ptr()),
+                         reinterpret_cast<dt*>(y.data_ptr()), g.rows, g.C,
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         gamma_pf, beta_pf, relu);
+    else if (add)
       hipLaunchKernelGGL((bn_fwd_norm_kernel<dt, true>), dim3(eblocks),
                          dim3(kBlock), 0, stream,
                          reinterpret_cast<const dt*>(x.data_ptr()),
