@@ -137,7 +137,11 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         # Per-collective HOST cost (async-enqueue launch-to-launch): paid
         # once per group regardless of device-side overlap — exactly what
         # merging amortizes at xGMI latencies. Folded into the solver's
-        # per-call constant.
+        # per-call constant. Note the serialized sweep's fitted alpha
+        # already contains ONE enqueue round-trip, so alpha + alpha_host
+        # slightly over-counts the per-call constant (~10 us); that bias
+        # is deliberate — it errs toward merging, the direction the
+        # unmodeled costs (hook python, RCCL channel setup) also point.
         a_host = prof.benchmark_host_overhead(dtype=self._comm_dtype)
         t = torch.tensor([a, b, a_host], dtype=torch.float64)
         if torch.cuda.is_available():
