@@ -69,3 +69,20 @@ class TestCompression:
             assert out.dtype == t.dtype
             tol = {'none': 0, 'fp16': 1e-3, 'bf16': 1e-2}[name]
             assert torch.allclose(out, t, atol=tol, rtol=tol)
+
+
+def test_benchmark_host_overhead_returns_positive():
+    """The per-collective host-cost probe (feeds the solver's launch
+    constant) must return a small positive seconds value even at
+    world=1 (noop handles: measures the python dispatch floor)."""
+    from mgwfbp_amd.profiling import CommunicationProfiler
+    import mgwfbp_amd.comm as comm
+    prof = CommunicationProfiler(comm.allreduce_async_, comm.synchronize)
+    t = prof.benchmark_host_overhead(num_calls=20, numel=1024)
+    assert 0 < t < 1e-2
+
+
+def test_sparse_cost_zero_numel():
+    from mgwfbp_amd import solver
+    assert solver.predict_sparse_allgather_time(1e-5, 1e-11, 0, 0.1,
+                                                8) == 0.0
